@@ -1,0 +1,249 @@
+#!/usr/bin/env python3
+"""Benchmark of the MI355X-native NeutronStarLite aggregation hot path.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W` (N>1 via
+torch.distributed.run, one rank per GPU over RCCL).  One "step" = one forward
+CSC aggregation + one backward CSR aggregation of the GCN layer over the full
+graph (BASELINE.json metric: aggregated edges/sec; 2E edges per step).
+
+Default workload (fits one GPU) = BASELINE config #2: Reddit-scale synthetic
+power-law RMAT (a,b,c,d)=(0.57,0.19,0.19,0.05), V=232 965, E=114M (+V self
+loops), seed 7, feat=602, fp32, weights = 1/sqrt(outdeg*indeg), features
+seeded U(-1,1) (seed 42), random-init-equivalent synthetic data (no datasets
+on the box).  config #1 (Cora) is a parity-test case, not a bench line.
+
+Rank 0 prints ONE JSON line, including:
+  roofline     — the dominant kernel (forward CSC aggregation): ALGORITHMIC
+                 bytes/launch = E*(4f+8) + V*8f (SURVEY.md §8d byte model,
+                 stated in DESIGN.md) divided by that kernel's average launch
+                 duration measured live with HIP events on the launching
+                 stream (via the C-ABI's nts_stream_kernel_ns); peak = 8 TB/s
+                 HBM3E.  `traffic` = measured HBM bytes/launch from a separate
+                 rocprofv3 --pmc pass (profiles/), injected with
+                 --traffic-bytes-per-launch; null when not provided.
+  cpu_baseline — the CPU oracle (oracle/oracle.c, kind "port") timed on this
+                 box's host cores over a bounded sample of the same workload
+                 (~10-30 s of CPU work), rank 0 at N=1 only.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+
+def log(msg):
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def build_workload(args):
+    from neutronstarlite_amd import graph as G
+    t0 = time.time()
+    if args.graph == "reddit":
+        v, e = 232_965, 114_000_000
+    elif args.graph == "rmat26":
+        v, e = 1 << 26, 1_000_000_000
+    elif args.graph == "small":
+        v, e = 10_000, 200_000
+    else:
+        raise SystemExit(f"unknown graph {args.graph}")
+    edges = G.rmat_edges(v, e, seed=7)
+    outd, ind = G.degrees(edges, v)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    log(f"graph generated: V={v} E={len(edges)} in {time.time()-t0:.1f}s")
+    return v, edges, w
+
+
+def cpu_baseline_leg(chunks, f, target_sec=15.0):
+    """Time the CPU oracle (test-infrastructure restatement of the
+    reference's ForwardCPUfuseOp loops) on a bounded sample of the same
+    workload. Returns the cpu_baseline JSON object."""
+    import oracle
+    ch = chunks[0]
+    v = ch.dst_n
+    rng = np.random.default_rng(42)
+    # bounded sample: prefix of destinations covering ~target edge count;
+    # calibrate on a small probe first.
+    probe_d = max(1, min(v, 2000))
+    x = rng.uniform(-1, 1, size=(ch.src_n, f)).astype(np.float32)
+    t0 = time.time()
+    oracle.csc_forward(ch.column_offset[:probe_d + 1], ch.row_indices,
+                       ch.edge_weight_forward, x, ch.src_s, probe_d, f)
+    dt = max(time.time() - t0, 1e-3)
+    probe_edges = int(ch.column_offset[probe_d])
+    eps = probe_edges / dt  # edges/sec estimate
+    target_edges = int(eps * target_sec)
+    n_d = int(np.searchsorted(ch.column_offset, target_edges))
+    n_d = max(probe_d, min(v, n_d))
+    e_f = int(ch.column_offset[n_d])
+    t0 = time.time()
+    oracle.csc_forward(ch.column_offset[:n_d + 1], ch.row_indices,
+                       ch.edge_weight_forward, x, ch.src_s, n_d, f)
+    t_fwd = time.time() - t0
+    # backward leg on a source prefix of similar size
+    n_s = int(np.searchsorted(ch.row_offset, e_f))
+    n_s = max(1, min(ch.src_n, n_s))
+    e_b = int(ch.row_offset[n_s])
+    g = rng.uniform(-1, 1, size=(v, f)).astype(np.float32)
+    t0 = time.time()
+    oracle.csr_backward(ch.row_offset[:n_s + 1], ch.column_indices,
+                        ch.edge_weight_backward, g, ch.dst_s, n_s, f)
+    t_bwd = time.time() - t0
+    value = (e_f + e_b) / (t_fwd + t_bwd)
+    return {
+        "value": round(value, 1), "unit": "aggregated_edges_per_sec",
+        "cores": oracle.num_threads(), "kind": "port",
+        "sample": (f"fwd {e_f} edges ({n_d} dsts) + bwd {e_b} edges "
+                   f"({n_s} srcs) of the f={f} workload, "
+                   f"{t_fwd + t_bwd:.1f}s on host cores"),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--feat", type=int, default=602)
+    ap.add_argument("--graph", default="reddit",
+                    choices=["reddit", "rmat26", "small"])
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--traffic-bytes-per-launch", type=float, default=None,
+                    help="measured HBM bytes per forward launch from a "
+                         "rocprofv3 --pmc pass (see profiles/)")
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+    from neutronstarlite_amd import graph as G, shim
+    from neutronstarlite_amd.ops import DeviceChunk, HipEngine
+    from neutronstarlite_amd.ring import RingGraph, ring_backward, ring_forward
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n = max(args.gpus, world)
+    distributed = world > 1
+    if args.gpus > 1 and world == 1:
+        raise SystemExit("--gpus N>1 must be launched via torch.distributed.run")
+
+    assert torch.cuda.is_available(), "bench needs an MI355X"
+    torch.cuda.set_device(local_rank)
+    dev = torch.device("cuda", local_rank)
+    if distributed:
+        dist.init_process_group("nccl")
+
+    f = args.feat
+    # every rank generates the same seeded graph, so partition offsets agree
+    v, edges, w = build_workload(args)
+    if distributed:
+        offs = G.partition_offsets(edges, v, world)
+    else:
+        offs = np.array([0, v], dtype=np.uint32)
+    t0 = time.time()
+    chunks = G.build_chunks(edges, w, offs, rank)
+    log(f"rank {rank}: chunks built in {time.time()-t0:.1f}s")
+
+    e_total = len(edges)
+    lo, hi = int(offs[rank]), int(offs[rank + 1])
+    dchunks = [DeviceChunk(ch, dev) for ch in chunks]
+    engine = HipEngine()
+    engine.stream.timing(True)
+    rg = RingGraph(offs, rank, dchunks, dev)
+
+    rng = np.random.default_rng(42 + rank)
+    x = torch.from_numpy(
+        rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
+    gy = torch.from_numpy(
+        rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
+
+    def step():
+        y = ring_forward(rg, x, engine)
+        gx = ring_backward(rg, gy, engine)
+        return y, gx
+
+    log(f"rank {rank}/{n}: warmup {args.warmup}")
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    engine.stream.timing_reset()
+
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.time() - t0
+    if distributed:
+        t = torch.tensor([elapsed], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    # roofline for the dominant (forward) kernel, from HIP events on the
+    # launching stream; at N=1 there is exactly one fwd launch per step.
+    fwd_ns = engine.stream.kernel_ns(shim.KTAG_FWD)
+    fwd_launches = engine.stream.kernel_launches(shim.KTAG_FWD)
+    local_edges = sum(ch.edge_size for ch in chunks)
+    # algorithmic bytes per launch: per edge f floats of source
+    # row + u32 index + f32 weight; per output row one read + one write.
+    edges_per_launch = local_edges / max(1, len(chunks))
+    algo_bytes_launch = (edges_per_launch * (4 * f + 8)
+                         + (hi - lo) * 8 * f / max(1, len(chunks)))
+    avg_launch_ns = fwd_ns / max(1, fwd_launches)
+    achieved = algo_bytes_launch / max(avg_launch_ns, 1e-9)  # GB/s
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved, 1),
+        "peak": 8000.0,
+        "unit": "GB/s",
+        "frac": round(achieved / 8000.0, 4),
+        "traffic": args.traffic_bytes_per_launch,
+    }
+
+    cpu_baseline = None
+    if rank == 0 and n == 1 and not args.no_cpu_baseline:
+        log("cpu baseline (oracle, bounded sample)")
+        cpu_baseline = cpu_baseline_leg(chunks, f)
+
+    value = args.steps * 2.0 * e_total / elapsed  # fwd + bwd edges, all ranks
+    if rank == 0:
+        out = {
+            "metric": "aggregated_edges_per_sec",
+            "value": round(value, 1),
+            "unit": "edges/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "f32",
+            "data": "synthetic",
+            "config": {
+                "workload": (f"{args.graph}: RMAT V={v} E={e_total} "
+                             f"feat={f} GCN-layer aggregation fwd+bwd "
+                             "(BASELINE config #2)" if args.graph == "reddit"
+                             else f"{args.graph}: RMAT V={v} E={e_total} feat={f}"),
+                "V": v, "E": e_total, "feat": f,
+                "parallelism": f"graph-partitioned dp{n}",
+                "edges_per_step": 2 * e_total,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,741 @@
+/* nts_hip.hip — MI355X-native (gfx950) kernels + C-ABI shim for the
+ * NeutronStarLite neighbor-aggregation hot path.
+ *
+ * Implements include/nts_hip.h.  This is a from-scratch CDNA4 design, not a
+ * port of the reference's CUDA kernels (/root/reference/cuda/*.cuh semantics
+ * are cited in the header; their fixed <<<128,512>>> geometry, LDS-atomic
+ * staging and f<=512 dispatch split are deliberately NOT reproduced):
+ *
+ *  - The aggregation is HBM-bandwidth-bound gather/scatter (no dense
+ *    contraction), so the design maximizes coalesced row traffic: one
+ *    64-lane wavefront group reads a source row slab per edge as wide
+ *    vector loads (dwordx4 / dwordx2 / 4x strided dword chosen from the
+ *    feature width and pointer alignment) and accumulates in REGISTERS —
+ *    no LDS staging, no per-element shared-memory atomics.
+ *  - Power-law load balance: columns are decomposed on device into bounded
+ *    work items of <= NTS_SPLIT consecutive edges of one vertex (cached per
+ *    static chunk); single-item vertices do plain read-modify-write stores,
+ *    split (hub) vertices merge with device-scope fp32 atomics.
+ *  - Launches are grid-stride with ~2048 blocks of 256 threads (4 waves),
+ *    plenty to fill 256 CUs across 8 XCDs; occupancy (8 waves/SIMD at this
+ *    register budget) hides HBM latency.
+ *
+ * Verified against the CPU oracle (oracle/oracle.c) in tests/test_gpu_parity.py.
+ */
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+#include <utility>
+#include <vector>
+
+#include "../../include/nts_hip.h"
+
+#define NTS_CHECK(cmd)                                                        \
+  do {                                                                        \
+    hipError_t e_ = (cmd);                                                    \
+    if (e_ != hipSuccess) {                                                   \
+      fprintf(stderr, "nts_hip: %s failed: %s (%s:%d)\n", #cmd,               \
+              hipGetErrorString(e_), __FILE__, __LINE__);                     \
+      abort(); /* reference CHECK culture, ntsCUDAGraphOP.cu:13-19 */         \
+    }                                                                         \
+  } while (0)
+
+namespace {
+constexpr uint32_t NTS_SPLIT = 256;   /* max edges per work item */
+constexpr int NTS_BLOCK = 256;        /* 4 waves */
+constexpr int NTS_MAX_BLOCKS = 2048;  /* 8 blocks/CU on 256 CUs */
+
+struct ItemsBuf {
+  uint4 *items = nullptr;
+  uint32_t *counter = nullptr;  /* device: n_items after build */
+  uint64_t cap = 0;
+};
+}  // namespace
+
+struct nts_stream {
+  hipStream_t stream = nullptr;
+  bool owned = false;
+  bool timing = false;
+  struct Rec { hipEvent_t a, b; int tag; };
+  std::vector<Rec> pending;
+  std::vector<std::pair<hipEvent_t, hipEvent_t>> freeev;
+  double acc_ns[NTS_KTAG_COUNT] = {};
+  long long acc_n[NTS_KTAG_COUNT] = {};
+  std::map<std::pair<const void *, uint32_t>, ItemsBuf> items_cache;
+};
+
+namespace {
+struct Tic {
+  nts_stream *s;
+  int tag;
+  hipEvent_t a = nullptr, b = nullptr;
+  Tic(nts_stream *s_, int tag_) : s(s_), tag(tag_) {
+    if (!s->timing) return;
+    if (!s->freeev.empty()) {
+      a = s->freeev.back().first;
+      b = s->freeev.back().second;
+      s->freeev.pop_back();
+    } else {
+      NTS_CHECK(hipEventCreate(&a));
+      NTS_CHECK(hipEventCreate(&b));
+    }
+    NTS_CHECK(hipEventRecord(a, s->stream));
+  }
+  ~Tic() {
+    if (!s->timing || !a) return;
+    NTS_CHECK(hipEventRecord(b, s->stream));
+    s->pending.push_back({a, b, tag});
+  }
+};
+
+void drain_timing(nts_stream *s) {
+  if (s->pending.empty()) return;
+  NTS_CHECK(hipStreamSynchronize(s->stream));
+  for (auto &r : s->pending) {
+    float ms = 0.f;
+    NTS_CHECK(hipEventElapsedTime(&ms, r.a, r.b));
+    s->acc_ns[r.tag] += (double)ms * 1e6;
+    s->acc_n[r.tag] += 1;
+    s->freeev.push_back({r.a, r.b});
+  }
+  s->pending.clear();
+}
+
+/* ------------------------------------------------------------------ */
+/* Work-item decomposition: one item = (vertex | shared-flag, first    */
+/* edge, edge count <= NTS_SPLIT).  Built on device, no host sync.     */
+/* ------------------------------------------------------------------ */
+__global__ void k_build_items(const uint32_t *__restrict__ offset,
+                              uint32_t batch, uint32_t split,
+                              uint4 *__restrict__ items,
+                              uint32_t *__restrict__ counter) {
+  for (uint32_t v = blockIdx.x * blockDim.x + threadIdx.x; v < batch;
+       v += gridDim.x * blockDim.x) {
+    const uint32_t e0 = offset[v];
+    const uint32_t deg = offset[v + 1] - e0;
+    if (!deg) continue;
+    const uint32_t n = (deg + split - 1) / split;
+    const uint32_t base = atomicAdd(counter, n);
+    const uint32_t flag = (n > 1) ? 0x80000000u : 0u;
+    for (uint32_t j = 0; j < n; ++j) {
+      items[base + j] = make_uint4(v | flag, e0 + j * split,
+                                   min(split, deg - j * split), 0u);
+    }
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* THE aggregation kernel.  One lane GROUP of G<=64 lanes processes    */
+/* one (work item, feature slab): loops the item's edges, vector-loads */
+/* the neighbor row slab coalesced, fma-accumulates in registers, then */
+/* stores once (RMW for exclusive vertices, atomicAdd for split hubs). */
+/* ELEM = vector width per lane: 4 (dwordx4), 2 (dwordx2), or 1 with   */
+/* NSTR=4 strided dwords (handles any f and any 4-byte alignment).     */
+/* Forward CSC:  nbr = row_indices   (global src),  out rows = dst.    */
+/* Backward CSR: nbr = column_indices (global dst), out rows = src.    */
+/* ------------------------------------------------------------------ */
+template <int ELEM, bool WITH_W>
+__global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
+    const uint4 *__restrict__ items, const uint32_t *__restrict__ n_items_p,
+    const uint32_t *__restrict__ nbr, const float *__restrict__ ew,
+    const float *__restrict__ in, float *__restrict__ out, uint32_t nbr_start,
+    uint32_t f, uint32_t G, uint32_t n_slabs) {
+  constexpr int NSTR = (ELEM == 1) ? 4 : 1;  /* strided sub-elements */
+  const uint32_t n_items = *n_items_p;
+  const uint64_t total = (uint64_t)n_items * n_slabs;
+  const uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t glane = threadIdx.x & (G - 1);
+  const uint32_t n_groups = (gridDim.x * blockDim.x) / G;
+  const uint32_t slabf = G * ELEM * NSTR;
+
+  for (uint64_t t = tid / G; t < total; t += n_groups) {
+    const uint32_t it = (uint32_t)(t / n_slabs);
+    const uint32_t slab = (uint32_t)(t % n_slabs);
+    const uint4 itm = items[it];
+    const uint32_t v = itm.x & 0x7fffffffu;
+    const bool shared_v = (itm.x >> 31) != 0;
+    const uint32_t e0 = itm.y, cnt = itm.z;
+    const uint32_t fbase = slab * slabf;
+
+    float acc[ELEM * NSTR] = {};
+    if (ELEM == 1) {
+      /* strided dword path: element j at fbase + glane + j*G */
+      bool any = fbase + glane < f;
+      if (any) {
+        for (uint32_t e = e0; e < e0 + cnt; ++e) {
+          const uint64_t src = nbr[e] - nbr_start;
+          const float w = WITH_W ? ew[e] : 1.0f;
+          const float *p = in + src * f + fbase + glane;
+#pragma unroll
+          for (int j = 0; j < NSTR; ++j)
+            if (fbase + glane + j * G < f) acc[j] = fmaf(w, p[j * G], acc[j]);
+        }
+        float *o = out + (uint64_t)v * f + fbase + glane;
+        if (!shared_v) {
+#pragma unroll
+          for (int j = 0; j < NSTR; ++j)
+            if (fbase + glane + j * G < f) o[j * G] += acc[j];
+        } else {
+#pragma unroll
+          for (int j = 0; j < NSTR; ++j)
+            if (fbase + glane + j * G < f) atomicAdd(&o[j * G], acc[j]);
+        }
+      }
+    } else {
+      const uint32_t off = fbase + glane * ELEM;
+      const int nv = (off + ELEM <= f) ? ELEM : (off < f ? (int)(f - off) : 0);
+      if (nv == ELEM) {
+        for (uint32_t e = e0; e < e0 + cnt; ++e) {
+          const uint64_t src = nbr[e] - nbr_start;
+          const float w = WITH_W ? ew[e] : 1.0f;
+          const float *p = in + src * f + off;
+          if (ELEM == 4) {
+            const float4 x = *reinterpret_cast<const float4 *>(p);
+            acc[0] = fmaf(w, x.x, acc[0]);
+            acc[1] = fmaf(w, x.y, acc[1]);
+            acc[2] = fmaf(w, x.z, acc[2]);
+            acc[3] = fmaf(w, x.w, acc[3]);
+          } else {
+            const float2 x = *reinterpret_cast<const float2 *>(p);
+            acc[0] = fmaf(w, x.x, acc[0]);
+            acc[1] = fmaf(w, x.y, acc[1]);
+          }
+        }
+      } else if (nv > 0) {
+        for (uint32_t e = e0; e < e0 + cnt; ++e) {
+          const uint64_t src = nbr[e] - nbr_start;
+          const float w = WITH_W ? ew[e] : 1.0f;
+          const float *p = in + src * f + off;
+          for (int j = 0; j < nv; ++j) acc[j] = fmaf(w, p[j], acc[j]);
+        }
+      }
+      if (nv > 0) {
+        float *o = out + (uint64_t)v * f + off;
+        if (!shared_v) {
+          if (nv == 4) {
+            float4 y = *reinterpret_cast<float4 *>(o);
+            y.x += acc[0]; y.y += acc[1]; y.z += acc[2]; y.w += acc[3];
+            *reinterpret_cast<float4 *>(o) = y;
+          } else if (nv == 2 && ELEM == 2) {
+            float2 y = *reinterpret_cast<float2 *>(o);
+            y.x += acc[0]; y.y += acc[1];
+            *reinterpret_cast<float2 *>(o) = y;
+          } else {
+            for (int j = 0; j < nv; ++j) o[j] += acc[j];
+          }
+        } else {
+          for (int j = 0; j < nv; ++j) atomicAdd(&o[j], acc[j]);
+        }
+      }
+    }
+  }
+}
+
+/* message unpack: records [u32 vid | f x f32], stride f+1 floats */
+__global__ void k_deserialize(const float *__restrict__ msg, uint32_t count,
+                              uint32_t part_start, float *__restrict__ dense,
+                              uint32_t f) {
+  const uint64_t total = (uint64_t)count * f;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < total;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    const uint64_t k = i / f;
+    const uint32_t r = (uint32_t)(i - k * f);
+    const float *rec = msg + k * (f + 1);
+    uint32_t vid = *reinterpret_cast<const uint32_t *>(rec);
+    dense[(uint64_t)(vid - part_start) * f + r] = rec[1 + r];
+  }
+}
+
+/* partial-sum merge: master[vid-part_start,:] += rec[1:] */
+__global__ void k_agg_msg(float *__restrict__ master,
+                          const float *__restrict__ msg, uint32_t count,
+                          uint32_t part_start, uint32_t f) {
+  const uint64_t total = (uint64_t)count * f;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < total;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    const uint64_t k = i / f;
+    const uint32_t r = (uint32_t)(i - k * f);
+    const float *rec = msg + k * (f + 1);
+    uint32_t vid = *reinterpret_cast<const uint32_t *>(rec);
+    atomicAdd(&master[(uint64_t)(vid - part_start) * f + r], rec[1 + r]);
+  }
+}
+
+/* dense-row pack/unpack for the RCCL ring (indices static, payload dense) */
+enum RowOp { ROW_GATHER, ROW_SCATTER, ROW_SCATTER_ADD };
+template <RowOp OP>
+__global__ void k_rows(float *__restrict__ dense, float *__restrict__ packed,
+                       const uint32_t *__restrict__ index, uint32_t count,
+                       uint32_t row_start, uint32_t f) {
+  const uint64_t total = (uint64_t)count * f;
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < total;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    const uint64_t k = i / f;
+    const uint32_t r = (uint32_t)(i - k * f);
+    const uint64_t d = (uint64_t)(index[k] - row_start) * f + r;
+    if (OP == ROW_GATHER) packed[i] = dense[d];
+    else if (OP == ROW_SCATTER) dense[d] = packed[i];
+    else atomicAdd(&dense[d], packed[i]);
+  }
+}
+
+/* ---------------- edge-wise (GAT) kernels ---------------- */
+/* wave-per-destination; lane-strided over the column's edges x f slots */
+enum EdgeOp { E_SCATTER_SRC, E_GATHER_SRC, E_SCATTER_DST, E_GATHER_DST,
+              E_SCATTER_GRAD };
+template <EdgeOp OP>
+__global__ void k_edge_op(float *__restrict__ message,
+                          float *__restrict__ vertex_feat,
+                          const uint32_t *__restrict__ row_indices,
+                          const uint32_t *__restrict__ column_offset,
+                          const uint32_t *__restrict__ mirror_index,
+                          uint32_t batch, uint32_t f) {
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t d = wave; d < batch; d += n_waves) {
+    const uint32_t e0 = column_offset[d], e1 = column_offset[d + 1];
+    const uint64_t nel = (uint64_t)(e1 - e0) * f;
+    for (uint64_t i = lane; i < nel; i += 64) {
+      const uint32_t e = e0 + (uint32_t)(i / f);
+      const uint32_t r = (uint32_t)(i % f);
+      uint64_t vrow;
+      if (OP == E_SCATTER_SRC || OP == E_GATHER_SRC)
+        vrow = (uint64_t)mirror_index[row_indices[e]] * f + r;
+      else
+        vrow = (uint64_t)d * f + r;
+      const uint64_t m = (uint64_t)e * f + r;
+      if (OP == E_SCATTER_SRC || OP == E_SCATTER_DST) message[m] = vertex_feat[vrow];
+      else if (OP == E_SCATTER_GRAD) atomicAdd(&message[m], vertex_feat[vrow]);
+      else atomicAdd(&vertex_feat[vrow], message[m]);
+    }
+  }
+}
+
+/* per-dst softmax over incident edge values, per feature slot.
+ * Restates edge_softmax_forward_block (ntsCUDADistKernel.cuh:166-213):
+ * out[e] = exp(in[e]) / sum_{e' in dst} exp(in[e']), no max subtraction
+ * (the exercised GAT config clamps attention scores with leaky_relu before
+ * this, keeping exp in range); cached = out for the backward pass. */
+template <bool BACKWARD>
+__global__ void k_edge_softmax(float *__restrict__ out,
+                               const float *__restrict__ in,
+                               const float *__restrict__ cached,
+                               const uint32_t *__restrict__ column_offset,
+                               uint32_t batch, uint32_t f) {
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t d = wave; d < batch; d += n_waves) {
+    const uint32_t e0 = column_offset[d], e1 = column_offset[d + 1];
+    for (uint32_t r = 0; r < f; ++r) {
+      float sum = 0.f;
+      for (uint32_t e = e0 + lane; e < e1; e += 64) {
+        const uint64_t m = (uint64_t)e * f + r;
+        if (BACKWARD) sum += in[m] * cached[m];
+        else sum += __expf(in[m]);
+      }
+#pragma unroll
+      for (int w = 32; w >= 1; w >>= 1) sum += __shfl_xor(sum, w, 64);
+      for (uint32_t e = e0 + lane; e < e1; e += 64) {
+        const uint64_t m = (uint64_t)e * f + r;
+        if (BACKWARD)
+          out[m] = in[m] * cached[m] - sum * cached[m];
+        else
+          out[m] = __expf(in[m]) / sum;
+      }
+    }
+  }
+}
+
+/* forward softmax also fills msg_cached */
+__global__ void k_copy(float *__restrict__ dst, const float *__restrict__ src,
+                       uint64_t n) {
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x)
+    dst[i] = src[i];
+}
+
+uint32_t grid_for(uint64_t threads) {
+  uint64_t b = (threads + NTS_BLOCK - 1) / NTS_BLOCK;
+  if (b < 1) b = 1;
+  if (b > NTS_MAX_BLOCKS) b = NTS_MAX_BLOCKS;
+  return (uint32_t)b;
+}
+
+ItemsBuf &get_items(nts_stream *s, const uint32_t *offset, uint32_t batch,
+                    uint32_t edges) {
+  auto key = std::make_pair((const void *)offset, batch);
+  auto it = s->items_cache.find(key);
+  if (it != s->items_cache.end()) return it->second;
+  ItemsBuf ib;
+  ib.cap = (uint64_t)batch + edges / NTS_SPLIT + 1;
+  NTS_CHECK(hipMalloc(&ib.items, ib.cap * sizeof(uint4)));
+  NTS_CHECK(hipMalloc(&ib.counter, sizeof(uint32_t)));
+  NTS_CHECK(hipMemsetAsync(ib.counter, 0, sizeof(uint32_t), s->stream));
+  {
+    Tic t(s, NTS_KTAG_ITEMS);
+    hipLaunchKernelGGL(k_build_items, dim3(grid_for(batch)), dim3(NTS_BLOCK), 0,
+                       s->stream, offset, batch, NTS_SPLIT, ib.items,
+                       ib.counter);
+  }
+  auto res = s->items_cache.emplace(key, ib);
+  return res.first->second;
+}
+
+void launch_gather(nts_stream *s, const float *in, float *out, const float *ew,
+                   const uint32_t *nbr, const uint32_t *offset,
+                   uint32_t nbr_start, uint32_t batch, uint32_t edges,
+                   uint32_t f, int with_weight, int tag) {
+  if (!batch || !edges || !f) return;
+  ItemsBuf &ib = get_items(s, offset, batch, edges);
+  const uintptr_t a = (uintptr_t)in | (uintptr_t)out;
+  int elem;
+  if (f % 4 == 0 && a % 16 == 0) elem = 4;
+  else if (f % 2 == 0 && a % 8 == 0) elem = 2;
+  else elem = 1;
+  const int nstr = (elem == 1) ? 4 : 1;
+  /* lane group: smallest power of two in [16,64] covering the slab work */
+  const uint32_t need = (f + elem * nstr - 1) / (elem * nstr);
+  uint32_t G = 64;
+  while (G / 2 >= need && G > 16) G /= 2;
+  const uint32_t slabf = G * elem * nstr;
+  const uint32_t n_slabs = (f + slabf - 1) / slabf;
+  const uint64_t bound_groups = ((uint64_t)batch + edges / NTS_SPLIT + 1) * n_slabs;
+  const uint32_t grid = grid_for(bound_groups * G);
+  Tic t(s, tag);
+#define NTS_LAUNCH(E, W)                                                      \
+  hipLaunchKernelGGL((k_gather_spmm<E, W>), dim3(grid), dim3(NTS_BLOCK), 0,   \
+                     s->stream, ib.items, ib.counter, nbr, ew, in, out,       \
+                     nbr_start, f, G, n_slabs)
+  if (with_weight) {
+    if (elem == 4) NTS_LAUNCH(4, true);
+    else if (elem == 2) NTS_LAUNCH(2, true);
+    else NTS_LAUNCH(1, true);
+  } else {
+    if (elem == 4) NTS_LAUNCH(4, false);
+    else if (elem == 2) NTS_LAUNCH(2, false);
+    else NTS_LAUNCH(1, false);
+  }
+#undef NTS_LAUNCH
+}
+}  // namespace
+
+/* ================= C-ABI ================= */
+extern "C" {
+
+nts_stream *nts_stream_create(void) {
+  nts_stream *s = new nts_stream();
+  NTS_CHECK(hipStreamCreateWithFlags(&s->stream, hipStreamNonBlocking));
+  s->owned = true;
+  return s;
+}
+
+nts_stream *nts_stream_wrap(void *hip_stream) {
+  nts_stream *s = new nts_stream();
+  s->stream = (hipStream_t)hip_stream;
+  s->owned = false;
+  return s;
+}
+
+void nts_stream_destroy(nts_stream *s) {
+  if (!s) return;
+  drain_timing(s);
+  for (auto &p : s->freeev) {
+    hipEventDestroy(p.first);
+    hipEventDestroy(p.second);
+  }
+  for (auto &kv : s->items_cache) {
+    hipFree(kv.second.items);
+    hipFree(kv.second.counter);
+  }
+  if (s->owned) hipStreamDestroy(s->stream);
+  delete s;
+}
+
+void nts_stream_sync(nts_stream *s) { NTS_CHECK(hipStreamSynchronize(s->stream)); }
+void *nts_stream_handle(nts_stream *s) { return (void *)s->stream; }
+
+void nts_stream_timing(nts_stream *s, int enable) { s->timing = enable != 0; }
+
+void nts_stream_timing_reset(nts_stream *s) {
+  drain_timing(s);
+  memset(s->acc_ns, 0, sizeof(s->acc_ns));
+  memset(s->acc_n, 0, sizeof(s->acc_n));
+}
+
+double nts_stream_kernel_ns(nts_stream *s, int tag) {
+  drain_timing(s);
+  return (tag >= 0 && tag < NTS_KTAG_COUNT) ? s->acc_ns[tag] : 0.0;
+}
+
+long long nts_stream_kernel_launches(nts_stream *s, int tag) {
+  drain_timing(s);
+  return (tag >= 0 && tag < NTS_KTAG_COUNT) ? s->acc_n[tag] : 0;
+}
+
+void *nts_malloc_gpu(long bytes) {
+  void *p = nullptr;
+  NTS_CHECK(hipMalloc(&p, (size_t)bytes));
+  return p;
+}
+
+void *nts_malloc_pinned(long bytes) {
+  void *p = nullptr;
+  NTS_CHECK(hipHostMalloc(&p, (size_t)bytes, hipHostMallocMapped));
+  return p;
+}
+
+void *nts_get_device_pointer(void *pinned) {
+  void *d = nullptr;
+  NTS_CHECK(hipHostGetDevicePointer(&d, pinned, 0));
+  return d;
+}
+
+void nts_free_gpu(void *p) { NTS_CHECK(hipFree(p)); }
+void nts_free_host(void *p) { NTS_CHECK(hipHostFree(p)); }
+
+void nts_zero_buffer(nts_stream *s, float *d, long n) {
+  NTS_CHECK(hipMemsetAsync(d, 0, (size_t)n * sizeof(float), s->stream));
+}
+
+void nts_memcpy_h2d(nts_stream *s, void *d, const void *h, long bytes, int sync) {
+  NTS_CHECK(hipMemcpyAsync(d, h, (size_t)bytes, hipMemcpyHostToDevice, s->stream));
+  if (sync) NTS_CHECK(hipStreamSynchronize(s->stream));
+}
+
+void nts_memcpy_d2h(nts_stream *s, void *h, const void *d, long bytes, int sync) {
+  NTS_CHECK(hipMemcpyAsync(h, d, (size_t)bytes, hipMemcpyDeviceToHost, s->stream));
+  if (sync) NTS_CHECK(hipStreamSynchronize(s->stream));
+}
+
+void nts_gather_by_dst_from_src(nts_stream *s, const float *input,
+                                float *output, const float *weight_forward,
+                                const nts_vid *row_indices,
+                                const nts_vid *column_offset, nts_vid src_start,
+                                nts_vid src_end, nts_vid dst_start,
+                                nts_vid dst_end, nts_vid edges,
+                                nts_vid batch_size, nts_vid feature_size,
+                                int with_weight) {
+  (void)src_end; (void)dst_start; (void)dst_end;
+  launch_gather(s, input, output, weight_forward, row_indices, column_offset,
+                src_start, batch_size, edges, feature_size, with_weight,
+                NTS_KTAG_FWD);
+}
+
+void nts_gather_by_src_from_dst(nts_stream *s, const float *input,
+                                float *output, const float *weight_backward,
+                                const nts_vid *row_offset,
+                                const nts_vid *column_indices,
+                                nts_vid src_start, nts_vid src_end,
+                                nts_vid dst_start, nts_vid dst_end,
+                                nts_vid edges, nts_vid batch_size,
+                                nts_vid feature_size, int with_weight) {
+  (void)src_start; (void)src_end; (void)dst_end;
+  launch_gather(s, input, output, weight_backward, column_indices, row_offset,
+                dst_start, batch_size, edges, feature_size, with_weight,
+                NTS_KTAG_BWD);
+}
+
+void nts_items_cache_clear(nts_stream *s) {
+  NTS_CHECK(hipStreamSynchronize(s->stream));
+  for (auto &kv : s->items_cache) {
+    hipFree(kv.second.items);
+    hipFree(kv.second.counter);
+  }
+  s->items_cache.clear();
+}
+
+void nts_deserialize_to_gpu(nts_stream *s, float *gpu_buffer, const float *msg,
+                            nts_vid count, nts_vid feature_size,
+                            nts_vid partition_start, nts_vid partition_end,
+                            int sync) {
+  (void)partition_end;
+  if (count) {
+    Tic t(s, NTS_KTAG_DESER);
+    hipLaunchKernelGGL(k_deserialize,
+                       dim3(grid_for((uint64_t)count * feature_size)),
+                       dim3(NTS_BLOCK), 0, s->stream, msg, count,
+                       partition_start, gpu_buffer, feature_size);
+  }
+  if (sync) NTS_CHECK(hipStreamSynchronize(s->stream));
+}
+
+void nts_aggregate_comm_result(nts_stream *s, float *master, const float *msg,
+                               nts_vid count, nts_vid feature_size,
+                               nts_vid partition_start, nts_vid partition_end,
+                               int sync) {
+  (void)partition_end;
+  if (count) {
+    Tic t(s, NTS_KTAG_AGGMSG);
+    hipLaunchKernelGGL(k_agg_msg,
+                       dim3(grid_for((uint64_t)count * feature_size)),
+                       dim3(NTS_BLOCK), 0, s->stream, master, msg, count,
+                       partition_start, feature_size);
+  }
+  if (sync) NTS_CHECK(hipStreamSynchronize(s->stream));
+}
+
+void nts_gather_rows(nts_stream *s, const float *dense, float *packed,
+                     const nts_vid *index, nts_vid count, nts_vid row_start,
+                     nts_vid feature_size) {
+  if (!count) return;
+  hipLaunchKernelGGL((k_rows<ROW_GATHER>),
+                     dim3(grid_for((uint64_t)count * feature_size)),
+                     dim3(NTS_BLOCK), 0, s->stream, const_cast<float *>(dense),
+                     packed, index, count, row_start, feature_size);
+}
+
+void nts_scatter_rows(nts_stream *s, float *dense, const float *packed,
+                      const nts_vid *index, nts_vid count, nts_vid row_start,
+                      nts_vid feature_size) {
+  if (!count) return;
+  hipLaunchKernelGGL((k_rows<ROW_SCATTER>),
+                     dim3(grid_for((uint64_t)count * feature_size)),
+                     dim3(NTS_BLOCK), 0, s->stream, dense,
+                     const_cast<float *>(packed), index, count, row_start,
+                     feature_size);
+}
+
+void nts_scatter_add_rows(nts_stream *s, float *dense, const float *packed,
+                          const nts_vid *index, nts_vid count,
+                          nts_vid row_start, nts_vid feature_size) {
+  if (!count) return;
+  hipLaunchKernelGGL((k_rows<ROW_SCATTER_ADD>),
+                     dim3(grid_for((uint64_t)count * feature_size)),
+                     dim3(NTS_BLOCK), 0, s->stream, dense,
+                     const_cast<float *>(packed), index, count, row_start,
+                     feature_size);
+}
+
+static void launch_edge_op(nts_stream *s, EdgeOp op, float *message,
+                           float *vfeat, const nts_vid *row_indices,
+                           const nts_vid *column_offset,
+                           const nts_vid *mirror_index, nts_vid batch,
+                           nts_vid f) {
+  if (!batch || !f) return;
+  Tic t(s, NTS_KTAG_EDGE);
+  const uint32_t grid = grid_for((uint64_t)batch * 64);
+#define NTS_ELAUNCH(OP)                                                       \
+  hipLaunchKernelGGL((k_edge_op<OP>), dim3(grid), dim3(NTS_BLOCK), 0,         \
+                     s->stream, message, vfeat, row_indices, column_offset,   \
+                     mirror_index, batch, f)
+  switch (op) {
+    case E_SCATTER_SRC: NTS_ELAUNCH(E_SCATTER_SRC); break;
+    case E_GATHER_SRC: NTS_ELAUNCH(E_GATHER_SRC); break;
+    case E_SCATTER_DST: NTS_ELAUNCH(E_SCATTER_DST); break;
+    case E_GATHER_DST: NTS_ELAUNCH(E_GATHER_DST); break;
+    case E_SCATTER_GRAD: NTS_ELAUNCH(E_SCATTER_GRAD); break;
+  }
+#undef NTS_ELAUNCH
+}
+
+void nts_scatter_src_mirror_to_msg(nts_stream *s, float *message,
+                                   const float *src_mirror_feature,
+                                   const nts_vid *row_indices,
+                                   const nts_vid *column_offset,
+                                   const nts_vid *mirror_index,
+                                   nts_vid batch_size, nts_vid feature_size) {
+  launch_edge_op(s, E_SCATTER_SRC, message,
+                 const_cast<float *>(src_mirror_feature), row_indices,
+                 column_offset, mirror_index, batch_size, feature_size);
+}
+
+void nts_gather_msg_to_src_mirror(nts_stream *s, float *src_mirror_feature,
+                                  const float *message,
+                                  const nts_vid *row_indices,
+                                  const nts_vid *column_offset,
+                                  const nts_vid *mirror_index,
+                                  nts_vid batch_size, nts_vid feature_size) {
+  launch_edge_op(s, E_GATHER_SRC, const_cast<float *>(message),
+                 src_mirror_feature, row_indices, column_offset, mirror_index,
+                 batch_size, feature_size);
+}
+
+void nts_scatter_dst_to_msg(nts_stream *s, float *message,
+                            const float *dst_feature,
+                            const nts_vid *row_indices,
+                            const nts_vid *column_offset, nts_vid batch_size,
+                            nts_vid feature_size) {
+  launch_edge_op(s, E_SCATTER_DST, message, const_cast<float *>(dst_feature),
+                 row_indices, column_offset, nullptr, batch_size, feature_size);
+}
+
+void nts_gather_msg_to_dst(nts_stream *s, float *dst_feature,
+                           const float *message, const nts_vid *row_indices,
+                           const nts_vid *column_offset, nts_vid batch_size,
+                           nts_vid feature_size) {
+  launch_edge_op(s, E_GATHER_DST, const_cast<float *>(message), dst_feature,
+                 row_indices, column_offset, nullptr, batch_size, feature_size);
+}
+
+void nts_scatter_grad_back_to_message(nts_stream *s, const float *input_grad,
+                                      float *message_grad,
+                                      const nts_vid *row_indices,
+                                      const nts_vid *column_offset,
+                                      nts_vid batch_size,
+                                      nts_vid feature_size) {
+  launch_edge_op(s, E_SCATTER_GRAD, message_grad,
+                 const_cast<float *>(input_grad), row_indices, column_offset,
+                 nullptr, batch_size, feature_size);
+}
+
+void nts_edge_softmax_forward(nts_stream *s, float *msg_output,
+                              const float *msg_input, float *msg_cached,
+                              const nts_vid *row_indices,
+                              const nts_vid *column_offset, nts_vid batch_size,
+                              nts_vid feature_size) {
+  (void)row_indices;
+  if (!batch_size || !feature_size) return;
+  {
+    Tic t(s, NTS_KTAG_EDGE);
+    hipLaunchKernelGGL((k_edge_softmax<false>),
+                       dim3(grid_for((uint64_t)batch_size * 64)),
+                       dim3(NTS_BLOCK), 0, s->stream, msg_output, msg_input,
+                       nullptr, column_offset, batch_size, feature_size);
+  }
+  /* cache = output for backward (reference caches at
+   * ntsCUDADistKernel.cuh:210) — one copy kernel; edge count comes from the
+   * offset array's last entry which the host layer knows. */
+  if (msg_cached && msg_cached != msg_output) {
+    uint32_t last = 0;
+    NTS_CHECK(hipMemcpyAsync(&last, column_offset + batch_size,
+                             sizeof(uint32_t), hipMemcpyDeviceToHost,
+                             s->stream));
+    NTS_CHECK(hipStreamSynchronize(s->stream));
+    const uint64_t n_edges_total = (uint64_t)last * feature_size;
+    hipLaunchKernelGGL(k_copy, dim3(grid_for(n_edges_total)), dim3(NTS_BLOCK),
+                       0, s->stream, msg_cached, msg_output, n_edges_total);
+  }
+}
+
+void nts_edge_softmax_backward(nts_stream *s, float *msg_input_grad,
+                               const float *msg_output_grad,
+                               const float *msg_cached,
+                               const nts_vid *row_indices,
+                               const nts_vid *column_offset,
+                               nts_vid batch_size, nts_vid feature_size) {
+  (void)row_indices;
+  if (!batch_size || !feature_size) return;
+  Tic t(s, NTS_KTAG_EDGE);
+  hipLaunchKernelGGL((k_edge_softmax<true>),
+                     dim3(grid_for((uint64_t)batch_size * 64)),
+                     dim3(NTS_BLOCK), 0, s->stream, msg_input_grad,
+                     msg_output_grad, msg_cached, column_offset, batch_size,
+                     feature_size);
+}
+
+int nts_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+void nts_set_device(int dev) { NTS_CHECK(hipSetDevice(dev)); }
+
+const char *nts_build_arch(void) { return "gfx950"; }
+
+}  /* extern "C" */

@@ -1,0 +1,110 @@
+"""Operator layer mirroring the reference's fused-op surface on torch tensors.
+
+Python mirror of nts::op::ntsGraphOp's calling convention
+(/root/reference/core/ntsBaseOp.hpp:24-48): ops take and return new tensors;
+`forward` consumes the layer input, `backward` the output grad.  The C++
+header mirror for linking toolkits unchanged lives in include/ and
+INTEGRATION.md; this module is the host driver used by tests and bench.
+
+  SingleGPUFuseOp  ~ ForwardSingleGPUfuseOp
+                     (core/ntsSingleGPUFusedGraphOp.hpp:48-71, driving
+                      Graph::forward_single/backward_single, graph.hpp:3806-3855)
+  DistGPUFuseOp    ~ ForwardGPUfuseOp
+                     (core/ntsDistGPUFusedGraphOp.hpp:48-91, driving
+                      sync_compute_decoupled / compute_sync_decoupled) — the
+                      ring exchange lives in ring.py on RCCL over xGMI.
+
+All compute runs through the C-ABI HIP shim (shim.Stream) on the caller's
+torch stream; there is NO CPU fallback — shim raises if the extension is
+missing.
+"""
+import numpy as np
+import torch
+
+from . import shim
+from .graph import Chunk
+
+
+def _u32_cuda(a: np.ndarray, device) -> torch.Tensor:
+    """Upload a u32 numpy array as an int32 cuda tensor (same bit pattern)."""
+    assert a.dtype == np.uint32
+    return torch.from_numpy(a.view(np.int32)).to(device)
+
+
+class DeviceChunk:
+    """A Chunk's CSC+CSR+weights resident in HBM
+    (CSC_segment_pinned::CopyGraphToDevice equivalent,
+    /root/reference/core/GraphSegment.cpp:178-220)."""
+
+    def __init__(self, ch: Chunk, device):
+        self.src_s, self.src_e = ch.src_s, ch.src_e
+        self.dst_s, self.dst_e = ch.dst_s, ch.dst_e
+        self.edge_size = ch.edge_size
+        self.column_offset = _u32_cuda(ch.column_offset, device)
+        self.row_indices = _u32_cuda(ch.row_indices, device)
+        self.w_fwd = torch.from_numpy(ch.edge_weight_forward).to(device)
+        self.row_offset = _u32_cuda(ch.row_offset, device)
+        self.column_indices = _u32_cuda(ch.column_indices, device)
+        self.w_bwd = torch.from_numpy(ch.edge_weight_backward).to(device)
+
+    @property
+    def dst_n(self):
+        return self.dst_e - self.dst_s
+
+    @property
+    def src_n(self):
+        return self.src_e - self.src_s
+
+
+class HipEngine:
+    """The product aggregation engine: HIP kernels via the C-ABI on the
+    current torch stream."""
+
+    def __init__(self):
+        self.stream = shim.Stream.wrap_torch_current()
+
+    def csc_forward(self, ch: DeviceChunk, x_block: torch.Tensor,
+                    y: torch.Tensor, with_weight=True):
+        """y[0:dst_n] += CSC-aggregate of x_block (dense over ch's src range)."""
+        f = y.shape[1]
+        assert x_block.is_cuda and y.is_cuda and x_block.dtype == torch.float32
+        assert x_block.is_contiguous() and y.is_contiguous()
+        assert x_block.shape[0] == ch.src_n and y.shape[0] == ch.dst_n
+        self.stream.gather_by_dst_from_src(
+            x_block.data_ptr(), y.data_ptr(), ch.w_fwd.data_ptr(),
+            ch.row_indices.data_ptr(), ch.column_offset.data_ptr(),
+            ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
+            ch.edge_size, ch.dst_n, f, with_weight)
+
+    def csr_backward(self, ch: DeviceChunk, grad_block: torch.Tensor,
+                     out: torch.Tensor, with_weight=True):
+        """out[0:src_n] += CSR-aggregate of grad_block (dense over dst range)."""
+        f = out.shape[1]
+        assert grad_block.is_contiguous() and out.is_contiguous()
+        assert grad_block.shape[0] == ch.dst_n and out.shape[0] == ch.src_n
+        self.stream.gather_by_src_from_dst(
+            grad_block.data_ptr(), out.data_ptr(), ch.w_bwd.data_ptr(),
+            ch.row_offset.data_ptr(), ch.column_indices.data_ptr(),
+            ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
+            ch.edge_size, ch.src_n, f, with_weight)
+
+
+class SingleGPUFuseOp:
+    """ForwardSingleGPUfuseOp equivalent: whole graph as one chunk on one GPU
+    (core/ntsSingleGPUFusedGraphOp.hpp:48-71)."""
+
+    def __init__(self, dchunk: DeviceChunk, engine: HipEngine = None):
+        self.ch = dchunk
+        self.engine = engine or HipEngine()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y = torch.zeros(self.ch.dst_n, x.shape[1], dtype=torch.float32,
+                        device=x.device)
+        self.engine.csc_forward(self.ch, x, y)
+        return y
+
+    def backward(self, grad_y: torch.Tensor) -> torch.Tensor:
+        gx = torch.zeros(self.ch.src_n, grad_y.shape[1], dtype=torch.float32,
+                         device=grad_y.device)
+        self.engine.csr_backward(self.ch, grad_y, gx)
+        return gx

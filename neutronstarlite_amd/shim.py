@@ -1,0 +1,218 @@
+"""ctypes binding of the C-ABI in include/nts_hip.h (the product compute path).
+
+FAILS LOUDLY if the HIP extension is missing: there is no CPU fallback in the
+product path — the oracle under oracle/ is test infrastructure only.
+"""
+import ctypes
+import os
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libnts_hip.so")
+
+# kernel-time tags (enum nts_ktag in include/nts_hip.h)
+KTAG_FWD = 0
+KTAG_BWD = 1
+KTAG_DESER = 2
+KTAG_AGGMSG = 3
+KTAG_ITEMS = 4
+KTAG_EDGE = 5
+
+_c = ctypes
+_vp = _c.c_void_p
+_u32 = _c.c_uint32
+_i32 = _c.c_int
+_i64 = _c.c_long
+
+
+class NtsHipMissing(RuntimeError):
+    pass
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_SO):
+        raise NtsHipMissing(
+            f"HIP extension not built: {_SO} missing. Run "
+            "`python -m neutronstarlite_amd.build` (or __graft_entry__.build()). "
+            "The product path has no CPU fallback.")
+    l = ctypes.CDLL(_SO)
+    # streams / timing
+    l.nts_stream_create.restype = _vp
+    l.nts_stream_wrap.restype = _vp
+    l.nts_stream_wrap.argtypes = [_vp]
+    l.nts_stream_destroy.argtypes = [_vp]
+    l.nts_stream_sync.argtypes = [_vp]
+    l.nts_stream_handle.restype = _vp
+    l.nts_stream_handle.argtypes = [_vp]
+    l.nts_stream_timing.argtypes = [_vp, _i32]
+    l.nts_stream_timing_reset.argtypes = [_vp]
+    l.nts_stream_kernel_ns.restype = _c.c_double
+    l.nts_stream_kernel_ns.argtypes = [_vp, _i32]
+    l.nts_stream_kernel_launches.restype = _c.c_longlong
+    l.nts_stream_kernel_launches.argtypes = [_vp, _i32]
+    # memory
+    l.nts_malloc_gpu.restype = _vp
+    l.nts_malloc_gpu.argtypes = [_i64]
+    l.nts_malloc_pinned.restype = _vp
+    l.nts_malloc_pinned.argtypes = [_i64]
+    l.nts_get_device_pointer.restype = _vp
+    l.nts_get_device_pointer.argtypes = [_vp]
+    l.nts_free_gpu.argtypes = [_vp]
+    l.nts_free_host.argtypes = [_vp]
+    l.nts_zero_buffer.argtypes = [_vp, _vp, _i64]
+    l.nts_memcpy_h2d.argtypes = [_vp, _vp, _vp, _i64, _i32]
+    l.nts_memcpy_d2h.argtypes = [_vp, _vp, _vp, _i64, _i32]
+    # hot kernels
+    l.nts_gather_by_dst_from_src.argtypes = [_vp] + [_vp] * 5 + [_u32] * 7 + [_i32]
+    l.nts_gather_by_src_from_dst.argtypes = [_vp] + [_vp] * 5 + [_u32] * 7 + [_i32]
+    l.nts_items_cache_clear.argtypes = [_vp]
+    l.nts_deserialize_to_gpu.argtypes = [_vp, _vp, _vp, _u32, _u32, _u32, _u32, _i32]
+    l.nts_aggregate_comm_result.argtypes = [_vp, _vp, _vp, _u32, _u32, _u32, _u32, _i32]
+    l.nts_gather_rows.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32, _u32]
+    l.nts_scatter_rows.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32, _u32]
+    l.nts_scatter_add_rows.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32, _u32]
+    # edge kernels
+    l.nts_scatter_src_mirror_to_msg.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_gather_msg_to_src_mirror.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_scatter_dst_to_msg.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
+    l.nts_gather_msg_to_dst.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
+    l.nts_scatter_grad_back_to_message.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
+    l.nts_edge_softmax_forward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_device_count.restype = _i32
+    l.nts_set_device.argtypes = [_i32]
+    l.nts_build_arch.restype = _c.c_char_p
+    _lib = l
+    return l
+
+
+class Stream:
+    """Thin RAII wrapper over nts_stream (replaces Cuda_Stream,
+    /root/reference/cuda/ntsCUDA.hpp:97-217)."""
+
+    def __init__(self, wrap_hip_stream=None):
+        l = lib()
+        if wrap_hip_stream:
+            self.h = l.nts_stream_wrap(_vp(wrap_hip_stream))
+        else:
+            self.h = l.nts_stream_create()
+        self._lib = l
+
+    @classmethod
+    def wrap_torch_current(cls):
+        import torch
+        return cls(wrap_hip_stream=torch.cuda.current_stream().cuda_stream)
+
+    def sync(self):
+        self._lib.nts_stream_sync(self.h)
+
+    def timing(self, enable=True):
+        self._lib.nts_stream_timing(self.h, 1 if enable else 0)
+
+    def timing_reset(self):
+        self._lib.nts_stream_timing_reset(self.h)
+
+    def kernel_ns(self, tag):
+        return self._lib.nts_stream_kernel_ns(self.h, tag)
+
+    def kernel_launches(self, tag):
+        return self._lib.nts_stream_kernel_launches(self.h, tag)
+
+    def zero(self, dptr, n_floats):
+        self._lib.nts_zero_buffer(self.h, _vp(dptr), n_floats)
+
+    def gather_by_dst_from_src(self, x_ptr, y_ptr, w_ptr, row_indices_ptr,
+                               column_offset_ptr, src_s, src_e, dst_s, dst_e,
+                               edges, batch, f, with_weight=True):
+        self._lib.nts_gather_by_dst_from_src(
+            self.h, _vp(x_ptr), _vp(y_ptr), _vp(w_ptr), _vp(row_indices_ptr),
+            _vp(column_offset_ptr), src_s, src_e, dst_s, dst_e, edges, batch,
+            f, 1 if with_weight else 0)
+
+    def gather_by_src_from_dst(self, g_ptr, y_ptr, w_ptr, row_offset_ptr,
+                               column_indices_ptr, src_s, src_e, dst_s, dst_e,
+                               edges, batch, f, with_weight=True):
+        self._lib.nts_gather_by_src_from_dst(
+            self.h, _vp(g_ptr), _vp(y_ptr), _vp(w_ptr), _vp(row_offset_ptr),
+            _vp(column_indices_ptr), src_s, src_e, dst_s, dst_e, edges, batch,
+            f, 1 if with_weight else 0)
+
+    def deserialize_to_gpu(self, dense_ptr, msg_ptr, count, f, part_s, part_e,
+                           sync=False):
+        self._lib.nts_deserialize_to_gpu(self.h, _vp(dense_ptr), _vp(msg_ptr),
+                                         count, f, part_s, part_e,
+                                         1 if sync else 0)
+
+    def aggregate_comm_result(self, master_ptr, msg_ptr, count, f, part_s,
+                              part_e, sync=False):
+        self._lib.nts_aggregate_comm_result(self.h, _vp(master_ptr),
+                                            _vp(msg_ptr), count, f, part_s,
+                                            part_e, 1 if sync else 0)
+
+    def gather_rows(self, dense_ptr, packed_ptr, index_ptr, count, row_start, f):
+        self._lib.nts_gather_rows(self.h, _vp(dense_ptr), _vp(packed_ptr),
+                                  _vp(index_ptr), count, row_start, f)
+
+    def scatter_rows(self, dense_ptr, packed_ptr, index_ptr, count, row_start, f):
+        self._lib.nts_scatter_rows(self.h, _vp(dense_ptr), _vp(packed_ptr),
+                                   _vp(index_ptr), count, row_start, f)
+
+    def scatter_add_rows(self, dense_ptr, packed_ptr, index_ptr, count,
+                         row_start, f):
+        self._lib.nts_scatter_add_rows(self.h, _vp(dense_ptr), _vp(packed_ptr),
+                                       _vp(index_ptr), count, row_start, f)
+
+    def scatter_src_mirror_to_msg(self, msg, mirror_feat, row_indices,
+                                  column_offset, mirror_index, batch, f):
+        self._lib.nts_scatter_src_mirror_to_msg(
+            self.h, _vp(msg), _vp(mirror_feat), _vp(row_indices),
+            _vp(column_offset), _vp(mirror_index), batch, f)
+
+    def gather_msg_to_src_mirror(self, mirror_feat, msg, row_indices,
+                                 column_offset, mirror_index, batch, f):
+        self._lib.nts_gather_msg_to_src_mirror(
+            self.h, _vp(mirror_feat), _vp(msg), _vp(row_indices),
+            _vp(column_offset), _vp(mirror_index), batch, f)
+
+    def scatter_dst_to_msg(self, msg, dst_feat, row_indices, column_offset,
+                           batch, f):
+        self._lib.nts_scatter_dst_to_msg(self.h, _vp(msg), _vp(dst_feat),
+                                         _vp(row_indices), _vp(column_offset),
+                                         batch, f)
+
+    def gather_msg_to_dst(self, dst_feat, msg, row_indices, column_offset,
+                          batch, f):
+        self._lib.nts_gather_msg_to_dst(self.h, _vp(dst_feat), _vp(msg),
+                                        _vp(row_indices), _vp(column_offset),
+                                        batch, f)
+
+    def scatter_grad_back_to_message(self, input_grad, msg_grad, row_indices,
+                                     column_offset, batch, f):
+        self._lib.nts_scatter_grad_back_to_message(
+            self.h, _vp(input_grad), _vp(msg_grad), _vp(row_indices),
+            _vp(column_offset), batch, f)
+
+    def edge_softmax_forward(self, msg_out, msg_in, msg_cached, row_indices,
+                             column_offset, batch, f):
+        self._lib.nts_edge_softmax_forward(
+            self.h, _vp(msg_out), _vp(msg_in), _vp(msg_cached),
+            _vp(row_indices), _vp(column_offset), batch, f)
+
+    def edge_softmax_backward(self, msg_in_grad, msg_out_grad, msg_cached,
+                              row_indices, column_offset, batch, f):
+        self._lib.nts_edge_softmax_backward(
+            self.h, _vp(msg_in_grad), _vp(msg_out_grad), _vp(msg_cached),
+            _vp(row_indices), _vp(column_offset), batch, f)
+
+    def items_cache_clear(self):
+        self._lib.nts_items_cache_clear(self.h)
+
+    def destroy(self):
+        if self.h:
+            self._lib.nts_stream_destroy(self.h)
+            self.h = None

@@ -1,0 +1,156 @@
+/* oracle.c — TEST INFRASTRUCTURE ONLY.
+ *
+ * CPU restatement of NeutronStarLite's GNN neighbor-aggregation hot-path
+ * arithmetic, used exclusively as the parity oracle and as bench.py's
+ * `cpu_baseline` leg.  It is NOT the product: only `tests/`,
+ * `__graft_entry__.smoke()` and `bench.py`'s cpu_baseline may call into this
+ * library.  The product path is the HIP extension (neutronstarlite_amd/csrc)
+ * and must fail loudly when that extension is missing.
+ *
+ * Semantics restated from /root/reference (file:line cited per function):
+ *   - forward CSC aggregation:  ForwardCPUfuseOp::forward sparse_slot,
+ *     core/ntsCPUFusedGraphOp.hpp:81-106 (nts_comp axpy, core/ntsBaseOp.hpp:82-104);
+ *     GPU twin: aggregate_kernel_from_src_with_weight,
+ *     cuda/ntsCUDAFuseKernel.cuh:272-290.
+ *   - backward CSR aggregation: ForwardCPUfuseOp::backward,
+ *     core/ntsCPUFusedGraphOp.hpp:120-163; GPU twin:
+ *     aggregate_kernel_from_dst_with_weight, cuda/ntsCUDAFuseKernel.cuh:450-468.
+ *   - edge weight: nts_norm_degree = 1/(sqrt(outdeg(src))*sqrt(indeg(dst))),
+ *     core/ntsBaseOp.hpp:194-197, degrees of the loaded edge list clamped to
+ *     >=1, core/graph.hpp:4397-4401.
+ *   - message (de)serialization: record layout [u32 vid | f x f32],
+ *     comm/network.cpp:476-495 and cuda/ntsCUDATransferKernel.cuh:70-93;
+ *     partial-sum merge: aggregate_data_buffer_debug,
+ *     cuda/ntsCUDATransferKernel.cuh:49-68.
+ *
+ * Parity pinning: the reference repo ships no golden vectors or known-answer
+ * tests for this path (its only executable checks are structural,
+ * test/testcsr.cpp:40-45) — "parity unpinned" by the reference's own tests.
+ * We pin this oracle ourselves in tests/test_oracle.py against (a) the
+ * closed form Y[d] = sum_e w_e on all-ones features (the reference's own
+ * deterministic input convention, core/ntsDataloador.hpp:63-71), and
+ * (b) an independent float64 scipy.sparse implementation of the same
+ * (row, col, w) triplets, on the vendored Cora edge list
+ * (data/cora.2708.edge.self) and on seeded RMAT graphs.
+ *
+ * Accumulation order: ascending edge index within each destination (forward)
+ * / source (backward), matching the reference's sequential per-vertex loops;
+ * parallelism is across vertices only, so results are deterministic.
+ */
+
+#include <math.h>
+#include <stdint.h>
+#include <string.h>
+
+typedef uint32_t vid_t;     /* VertexId, dep/gemini/type.hpp:28 */
+typedef float val_t;        /* ValueType, dep/gemini/type.hpp:30 */
+
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* Degrees of the loaded edge list: edges are (src,dst) u32 pairs
+ * (Gemini binary format, data/reddit/note_for_input.txt).  Restates the
+ * degree accumulation of Graph::load_directed / generate_backward_structure
+ * (core/graph.hpp:1127+, 4203+) with the >=1 clamp of graph.hpp:4397-4401. */
+EXPORT void oracle_degrees(const vid_t *edges /* 2*E: src,dst pairs */,
+                           uint64_t e, vid_t v,
+                           vid_t *out_degree, vid_t *in_degree) {
+  memset(out_degree, 0, sizeof(vid_t) * v);
+  memset(in_degree, 0, sizeof(vid_t) * v);
+  for (uint64_t i = 0; i < e; i++) {
+    out_degree[edges[2 * i]] += 1;
+    in_degree[edges[2 * i + 1]] += 1;
+  }
+  for (vid_t i = 0; i < v; i++) {
+    if (out_degree[i] < 1) out_degree[i] = 1;
+    if (in_degree[i] < 1) in_degree[i] = 1;
+  }
+}
+
+/* nts_norm_degree per edge (core/ntsBaseOp.hpp:194-197). */
+EXPORT void oracle_norm_weights(const vid_t *src, const vid_t *dst, uint64_t e,
+                                const vid_t *out_degree, const vid_t *in_degree,
+                                val_t *w) {
+#pragma omp parallel for schedule(static)
+  for (int64_t i = 0; i < (int64_t)e; i++) {
+    w[i] = 1.0f / ((val_t)sqrt((double)out_degree[src[i]]) *
+                   (val_t)sqrt((double)in_degree[dst[i]]));
+  }
+}
+
+/* Forward CSC aggregation over one chunk:
+ *   out[d,:] += sum_{e in col(d)} w[e] * in[row_indices[e]-src_s,:]
+ * out is NOT zeroed here (caller zeroes, matching ForwardCPUfuseOp::forward's
+ * memset at ntsCPUFusedGraphOp.hpp:48).  column_offset is local over
+ * [0, dst_n]; row_indices hold global src ids (PartitionedGraph.hpp:399). */
+EXPORT void oracle_csc_forward(const vid_t *column_offset, const vid_t *row_indices,
+                               const val_t *weight, const val_t *in, val_t *out,
+                               vid_t src_s, vid_t dst_n, int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)dst_n; d++) {
+    val_t *od = out + d * f;
+    for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+      const val_t *src_row = in + (int64_t)(row_indices[e] - src_s) * f;
+      const val_t w = weight[e];
+      for (int64_t i = 0; i < f; i++) od[i] += src_row[i] * w; /* nts_comp */
+    }
+  }
+}
+
+/* Backward CSR aggregation over one chunk:
+ *   out[s,:] += sum_{e in row(s)} w[e] * grad[column_indices[e]-dst_s,:]
+ * (ForwardCPUfuseOp::backward pull loop, ntsCPUFusedGraphOp.hpp:131-145). */
+EXPORT void oracle_csr_backward(const vid_t *row_offset, const vid_t *column_indices,
+                                const val_t *weight, const val_t *grad, val_t *out,
+                                vid_t dst_s, vid_t src_n, int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t s = 0; s < (int64_t)src_n; s++) {
+    val_t *os = out + s * f;
+    for (vid_t e = row_offset[s]; e < row_offset[s + 1]; e++) {
+      const val_t *g_row = grad + (int64_t)(column_indices[e] - dst_s) * f;
+      const val_t w = weight[e];
+      for (int64_t i = 0; i < f; i++) os[i] += g_row[i] * w;
+    }
+  }
+}
+
+/* Unpack message records [u32 vid | f x f32] into dense rows vid-part_start
+ * (deSerializeToGPUkernel, cuda/ntsCUDATransferKernel.cuh:70-93; record
+ * stride is (f+1) floats, comm/network.cpp:476-495 / sizeofM). */
+EXPORT void oracle_deserialize(const val_t *msg, uint64_t count,
+                               vid_t part_start, val_t *dense, int64_t f) {
+#pragma omp parallel for schedule(static)
+  for (int64_t k = 0; k < (int64_t)count; k++) {
+    const val_t *rec = msg + k * (f + 1);
+    vid_t vid;
+    memcpy(&vid, rec, sizeof(vid_t));
+    memcpy(dense + (int64_t)(vid - part_start) * f, rec + 1, sizeof(val_t) * f);
+  }
+}
+
+/* Merge received partial gradient records into master rows:
+ *   master[vid-part_start,:] += rec[1:]
+ * (aggregate_data_buffer_debug, cuda/ntsCUDATransferKernel.cuh:49-68;
+ * the live kernel — the un-suffixed one at :30-47 is dead legacy). */
+EXPORT void oracle_agg_msg_to_master(val_t *master, const val_t *msg,
+                                     uint64_t count, vid_t part_start, int64_t f) {
+  /* sequential: reference merges one ring step at a time (graph.hpp:3601-3607) */
+  for (uint64_t k = 0; k < count; k++) {
+    const val_t *rec = msg + k * (f + 1);
+    vid_t vid;
+    memcpy(&vid, rec, sizeof(vid_t));
+    val_t *m = master + (int64_t)(vid - part_start) * f;
+    for (int64_t i = 0; i < f; i++) m[i] += rec[1 + i]; /* nts_acc */
+  }
+}
+
+EXPORT int oracle_num_threads(void) {
+#ifdef _OPENMP
+  return omp_get_max_threads();
+#else
+  return 1;
+#endif
+}

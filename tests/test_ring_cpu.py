@@ -1,0 +1,94 @@
+"""Multi-process ring-exchange logic under gloo (world_size 2, CPU): the
+distributed path of ring.py must reproduce the whole-graph oracle result.
+The aggregation engine is oracle-backed HERE ONLY (test infrastructure);
+the product engine is HIP (ops.HipEngine) and is covered by the gpu tests."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+import oracle
+from neutronstarlite_amd import graph as G
+from neutronstarlite_amd.ring import RingGraph, ring_backward, ring_forward
+
+V, E, F, SEED = 777, 9000, 11, 7
+WORLD = 2
+
+
+class OracleEngine:
+    """Test-only CPU engine with the engine protocol ring.py expects."""
+
+    def csc_forward(self, ch, x_block, y, with_weight=True):
+        assert with_weight
+        oracle.csc_forward(ch.column_offset, ch.row_indices,
+                           ch.edge_weight_forward, x_block.numpy(), ch.src_s,
+                           ch.dst_n, y.shape[1], out=y.numpy())
+
+    def csr_backward(self, ch, grad_block, out, with_weight=True):
+        assert with_weight
+        oracle.csr_backward(ch.row_offset, ch.column_indices,
+                            ch.edge_weight_backward, grad_block.numpy(),
+                            ch.dst_s, ch.src_n, out.shape[1], out=out.numpy())
+
+
+def _data():
+    edges = G.rmat_edges(V, E, seed=SEED)
+    outd, ind = G.degrees(edges, V)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    offs = G.partition_offsets(edges, V, WORLD)
+    rng = np.random.default_rng(42)
+    x = rng.uniform(-1, 1, size=(V, F)).astype(np.float32)
+    g = rng.uniform(-1, 1, size=(V, F)).astype(np.float32)
+    return edges, w, offs, x, g
+
+
+def _worker(rank, tmpdir, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        dist.init_process_group(
+            "gloo", init_method=f"file://{tmpdir}/pg", rank=rank,
+            world_size=WORLD)
+        edges, w, offs, x, g = _data()
+        chunks = G.build_chunks(edges, w, offs, rank)
+        rg = RingGraph(offs, rank, chunks, torch.device("cpu"))
+        lo, hi = int(offs[rank]), int(offs[rank + 1])
+        eng = OracleEngine()
+        y = ring_forward(rg, torch.from_numpy(x[lo:hi]).clone(), eng)
+        gx = ring_backward(rg, torch.from_numpy(g[lo:hi]).clone(), eng)
+        q.put((rank, y.numpy(), gx.numpy()))
+        dist.destroy_process_group()
+    except Exception as exc:  # surface worker failures to the main process
+        q.put((rank, "error", repr(exc)))
+        raise
+
+
+@pytest.mark.timeout(180)
+def test_ring_matches_whole_graph(tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker, args=(r, str(tmp_path), q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, y, gx = q.get()
+        assert not (isinstance(y, str) and y == "error"), f"rank {rank}: {gx}"
+        results[rank] = (y, gx)
+    for p in procs:
+        p.join(60)
+        assert p.exitcode == 0
+
+    edges, w, offs, x, g = _data()
+    ch = G.build_chunks(edges, w, np.array([0, V], dtype=np.uint32), 0)[0]
+    y_ref = oracle.csc_forward(ch.column_offset, ch.row_indices,
+                               ch.edge_weight_forward, x, 0, V, F)
+    gx_ref = oracle.csr_backward(ch.row_offset, ch.column_indices,
+                                 ch.edge_weight_backward, g, 0, V, F)
+    y_all = np.concatenate([results[r][0] for r in range(WORLD)])
+    gx_all = np.concatenate([results[r][1] for r in range(WORLD)])
+    assert np.allclose(y_all, y_ref, rtol=1e-4, atol=1e-5)
+    assert np.allclose(gx_all, gx_ref, rtol=1e-4, atol=1e-5)
