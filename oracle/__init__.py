@@ -37,6 +37,20 @@ def _load():
     lib.oracle_deserialize.argtypes = [f32p, ctypes.c_uint64, ctypes.c_uint32, f32p, ctypes.c_int64]
     lib.oracle_agg_msg_to_master.argtypes = [f32p, f32p, ctypes.c_uint64, ctypes.c_uint32, ctypes.c_int64]
     lib.oracle_num_threads.restype = ctypes.c_int
+    lib.oracle_scatter_src_to_msg.argtypes = [f32p, f32p, u32p, u32p, u32p,
+                                              ctypes.c_uint32, ctypes.c_int64]
+    lib.oracle_gather_msg_to_src.argtypes = [f32p, f32p, u32p, u32p, u32p,
+                                             ctypes.c_uint32, ctypes.c_int64]
+    lib.oracle_scatter_dst_to_msg.argtypes = [f32p, f32p, u32p,
+                                              ctypes.c_uint32, ctypes.c_int64]
+    lib.oracle_gather_msg_to_dst.argtypes = [f32p, f32p, u32p,
+                                             ctypes.c_uint32, ctypes.c_int64]
+    lib.oracle_scatter_grad_back_to_msg.argtypes = [f32p, f32p, u32p,
+                                                    ctypes.c_uint32, ctypes.c_int64]
+    lib.oracle_edge_softmax_forward.argtypes = [f32p, f32p, f32p, u32p,
+                                                ctypes.c_uint32, ctypes.c_int64]
+    lib.oracle_edge_softmax_backward.argtypes = [f32p, f32p, f32p, u32p,
+                                                 ctypes.c_uint32, ctypes.c_int64]
     _lib = lib
     return lib
 
@@ -101,3 +115,50 @@ def agg_msg_to_master(master, msg, count, part_start, f):
 
 def num_threads():
     return _load().oracle_num_threads()
+
+
+def scatter_src_to_msg(msg, mirror, row_indices, column_offset, mirror_index,
+                       batch, f):
+    _load().oracle_scatter_src_to_msg(_f32(msg), _f32(mirror),
+                                      _u32(row_indices), _u32(column_offset),
+                                      _u32(mirror_index), batch, f)
+    return msg
+
+
+def gather_msg_to_src(mirror, msg, row_indices, column_offset, mirror_index,
+                      batch, f):
+    _load().oracle_gather_msg_to_src(_f32(mirror), _f32(msg),
+                                     _u32(row_indices), _u32(column_offset),
+                                     _u32(mirror_index), batch, f)
+    return mirror
+
+
+def scatter_dst_to_msg(msg, dst_feat, column_offset, batch, f):
+    _load().oracle_scatter_dst_to_msg(_f32(msg), _f32(dst_feat),
+                                      _u32(column_offset), batch, f)
+    return msg
+
+
+def gather_msg_to_dst(dst_feat, msg, column_offset, batch, f):
+    _load().oracle_gather_msg_to_dst(_f32(dst_feat), _f32(msg),
+                                     _u32(column_offset), batch, f)
+    return dst_feat
+
+
+def scatter_grad_back_to_msg(input_grad, msg_grad, column_offset, batch, f):
+    _load().oracle_scatter_grad_back_to_msg(_f32(input_grad), _f32(msg_grad),
+                                            _u32(column_offset), batch, f)
+    return msg_grad
+
+
+def edge_softmax_forward(out, inp, cached, column_offset, batch, f):
+    _load().oracle_edge_softmax_forward(_f32(out), _f32(inp), _f32(cached),
+                                        _u32(column_offset), batch, f)
+    return out
+
+
+def edge_softmax_backward(in_grad, out_grad, cached, column_offset, batch, f):
+    _load().oracle_edge_softmax_backward(_f32(in_grad), _f32(out_grad),
+                                         _f32(cached), _u32(column_offset),
+                                         batch, f)
+    return in_grad

@@ -147,6 +147,120 @@ EXPORT void oracle_agg_msg_to_master(val_t *master, const val_t *msg,
   }
 }
 
+/* ---- GAT edge-valued path (config #5) ----
+ * Restated from the decomposed dist ops and their kernels:
+ *   scatter_src_mirror_to_msg / gather_msg_to_src_mirror / scatter_dst_to_msg
+ *   / gather_msg_to_dst (cuda/ntsCUDADistKernel.cuh:23-95, driven by
+ *   DistGPU*Op, core/ntsDistGPUGraphOp.hpp:145-300; CPU twins
+ *   core/ntsDistCPUGraphOp.hpp:127-242),
+ *   edge_softmax_forward_block / _backward_block
+ *   (cuda/ntsCUDADistKernel.cuh:166-260): per-destination softmax over
+ *   incident edge values WITHOUT max subtraction (exp at :192,209), cached
+ *   output; backward g_in[e] = s[e]*g_out[e] - s[e]*sum(s*g_out).
+ *   scatter_grad_back_to_messaage (cuda/ntsCUDAFuseKernel.cuh:492-506).
+ * The reference block kernels read one scalar per edge (f=1 semantics); we
+ * state the per-feature-slot generalization, identical at f=1. */
+
+EXPORT void oracle_scatter_src_to_msg(float *msg, const float *mirror,
+                                      const vid_t *row_indices,
+                                      const vid_t *column_offset,
+                                      const vid_t *mirror_index, vid_t batch,
+                                      int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+      const val_t *row = mirror + (int64_t)mirror_index[row_indices[e]] * f;
+      memcpy(msg + (int64_t)e * f, row, sizeof(val_t) * f);
+    }
+  }
+}
+
+EXPORT void oracle_gather_msg_to_src(float *mirror, const float *msg,
+                                     const vid_t *row_indices,
+                                     const vid_t *column_offset,
+                                     const vid_t *mirror_index, vid_t batch,
+                                     int64_t f) {
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+      val_t *row = mirror + (int64_t)mirror_index[row_indices[e]] * f;
+      const val_t *m = msg + (int64_t)e * f;
+      for (int64_t i = 0; i < f; i++) row[i] += m[i];
+    }
+  }
+}
+
+EXPORT void oracle_scatter_dst_to_msg(float *msg, const float *dst_feat,
+                                      const vid_t *column_offset, vid_t batch,
+                                      int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++)
+      memcpy(msg + (int64_t)e * f, dst_feat + d * f, sizeof(val_t) * f);
+  }
+}
+
+EXPORT void oracle_gather_msg_to_dst(float *dst_feat, const float *msg,
+                                     const vid_t *column_offset, vid_t batch,
+                                     int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+      const val_t *m = msg + (int64_t)e * f;
+      for (int64_t i = 0; i < f; i++) dst_feat[d * f + i] += m[i];
+    }
+  }
+}
+
+EXPORT void oracle_scatter_grad_back_to_msg(const float *input_grad,
+                                            float *msg_grad,
+                                            const vid_t *column_offset,
+                                            vid_t batch, int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+      float *m = msg_grad + (int64_t)e * f;
+      for (int64_t i = 0; i < f; i++) m[i] += input_grad[d * f + i];
+    }
+  }
+}
+
+EXPORT void oracle_edge_softmax_forward(float *out, const float *in,
+                                        float *cached,
+                                        const vid_t *column_offset,
+                                        vid_t batch, int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (int64_t r = 0; r < f; r++) {
+      double sum = 0.0; /* fp32-accumulated on GPU; fp64 here is the oracle */
+      for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++)
+        sum += expf(in[(int64_t)e * f + r]);
+      for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+        const float v = expf(in[(int64_t)e * f + r]) / (float)sum;
+        out[(int64_t)e * f + r] = v;
+        if (cached) cached[(int64_t)e * f + r] = v;
+      }
+    }
+  }
+}
+
+EXPORT void oracle_edge_softmax_backward(float *in_grad, const float *out_grad,
+                                         const float *cached,
+                                         const vid_t *column_offset,
+                                         vid_t batch, int64_t f) {
+#pragma omp parallel for schedule(dynamic, 64)
+  for (int64_t d = 0; d < (int64_t)batch; d++) {
+    for (int64_t r = 0; r < f; r++) {
+      double dot = 0.0;
+      for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++)
+        dot += out_grad[(int64_t)e * f + r] * cached[(int64_t)e * f + r];
+      for (vid_t e = column_offset[d]; e < column_offset[d + 1]; e++) {
+        const int64_t m = (int64_t)e * f + r;
+        in_grad[m] = out_grad[m] * cached[m] - (float)dot * cached[m];
+      }
+    }
+  }
+}
+
 EXPORT int oracle_num_threads(void) {
 #ifdef _OPENMP
   return omp_get_max_threads();
