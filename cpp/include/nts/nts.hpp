@@ -1,0 +1,236 @@
+/* nts.hpp — C++ host layer: the reference's operator surface over libtorch,
+ * computing through the C-ABI HIP shim (include/nts_hip.h).
+ *
+ * This keeps the calling convention of /root/reference's operator boundary
+ *  so model code written against it — e.g. the per-layer loop
+ * of toolkits/GCN.hpp:217-235 — compiles against this header:
+ *   - NtsVar = torch::Tensor (core/NtsScheduler.hpp:52)
+ *   - nts::op::ntsGraphOp: ctor (PartitionedGraph*, VertexSubset*),
+ *     NtsVar forward(NtsVar&), NtsVar backward(NtsVar&)
+ *     (core/ntsBaseOp.hpp:24-48)
+ *   - class names ForwardSingleGPUfuseOp / ForwardGPUfuseOp
+ *     (core/ntsSingleGPUFusedGraphOp.hpp:48-71, ntsDistGPUFusedGraphOp.hpp:48-91)
+ *   - NtsContext::runGraphOp<T> / appendNNOp / self_backward tape
+ *     (core/ntsContext.hpp:108-359)
+ * The implementations are new (MI355X-first; no host bounce, no MPI): only
+ * the surface matches.  Error culture: assert/abort, no exceptions
+ * (reference convention).
+ */
+#pragma once
+
+#include <torch/torch.h>
+
+#include <cassert>
+#include <cstdint>
+#include <functional>
+#include <vector>
+
+#include "nts_hip.h"
+
+namespace nts {
+
+using VertexId = uint32_t;   /* dep/gemini/type.hpp:28 */
+using ValueType = float;     /* dep/gemini/type.hpp:30 */
+using NtsVar = torch::Tensor;
+
+/* Active-vertex subset: carried for signature compatibility (full-batch
+ * configs run with all vertices active). */
+struct VertexSubset {
+  VertexId start = 0, end = 0;
+};
+
+/* One per-source-partition graph chunk resident in HBM: forward CSC +
+ * backward CSR + norm-degree weights (CSC_segment_pinned surface,
+ * core/GraphSegment.h:52-139; device upload ≙ CopyGraphToDevice,
+ * core/GraphSegment.cpp:178-220). */
+struct CSC_segment_pinned {
+  VertexId src_range[2] = {0, 0};
+  VertexId dst_range[2] = {0, 0};
+  VertexId edge_size = 0;
+  torch::Tensor column_offset;   // u32 as int32, device, [dst_n+1] local
+  torch::Tensor row_indices;     // global src ids, [E]
+  torch::Tensor edge_weight_forward;
+  torch::Tensor row_offset;      // [src_n+1] local
+  torch::Tensor column_indices;  // global dst ids, [E]
+  torch::Tensor edge_weight_backward;
+
+  VertexId dst_n() const { return dst_range[1] - dst_range[0]; }
+  VertexId src_n() const { return src_range[1] - src_range[0]; }
+
+  /* Build from host CSC+CSR arrays (u32 / f32), upload to `device`. */
+  static CSC_segment_pinned from_host(
+      VertexId src_s, VertexId src_e, VertexId dst_s, VertexId dst_e,
+      const uint32_t *col_off, const uint32_t *rows, const float *wf,
+      const uint32_t *row_off, const uint32_t *cols, const float *wb,
+      VertexId edges, torch::Device device) {
+    CSC_segment_pinned c;
+    c.src_range[0] = src_s; c.src_range[1] = src_e;
+    c.dst_range[0] = dst_s; c.dst_range[1] = dst_e;
+    c.edge_size = edges;
+    auto u32 = torch::TensorOptions().dtype(torch::kInt32);
+    auto f32 = torch::TensorOptions().dtype(torch::kFloat32);
+    auto up = [&](const void *p, int64_t n, torch::TensorOptions o) {
+      return torch::from_blob(const_cast<void *>(p), {n}, o).to(device);
+    };
+    c.column_offset = up(col_off, (int64_t)(dst_e - dst_s) + 1, u32);
+    c.row_indices = up(rows, edges, u32);
+    c.edge_weight_forward = up(wf, edges, f32);
+    c.row_offset = up(row_off, (int64_t)(src_e - src_s) + 1, u32);
+    c.column_indices = up(cols, edges, f32.dtype(torch::kInt32));
+    c.edge_weight_backward = up(wb, edges, f32);
+    return c;
+  }
+};
+
+/* Partitioned-graph view: per-source-partition chunks of THIS rank
+ * (PartitionedGraph surface, core/PartitionedGraph.hpp). */
+struct PartitionedGraph {
+  std::vector<CSC_segment_pinned *> graph_chunks;
+  int partition_id = 0;
+  std::vector<VertexId> partition_offset;  // [P+1]
+  nts_stream *stream = nullptr;            // compute stream (C-ABI)
+
+  PartitionedGraph() { stream = nts_stream_create(); }
+  ~PartitionedGraph() {
+    if (stream) nts_stream_destroy(stream);
+  }
+  VertexId owned_vertices() const {
+    return partition_offset[partition_id + 1] - partition_offset[partition_id];
+  }
+};
+
+namespace op {
+
+/* Abstract graph-op (core/ntsBaseOp.hpp:24-48 surface). */
+class ntsGraphOp {
+ public:
+  PartitionedGraph *partitioned_graph_ = nullptr;
+  VertexSubset *active_ = nullptr;
+  ntsGraphOp() {}
+  ntsGraphOp(PartitionedGraph *pg, VertexSubset *active)
+      : partitioned_graph_(pg), active_(active) {}
+  virtual ~ntsGraphOp() {}
+  virtual NtsVar forward(NtsVar &f_input) = 0;
+  virtual NtsVar backward(NtsVar &output_grad) = 0;
+};
+
+namespace detail {
+
+inline void csc_forward(nts_stream *s, const CSC_segment_pinned &c,
+                        const NtsVar &x, NtsVar &y) {
+  assert(x.is_cuda() && x.is_contiguous() &&
+         x.size(0) == (int64_t)c.src_n());
+  nts_gather_by_dst_from_src(
+      s, x.data_ptr<float>(), y.data_ptr<float>(),
+      c.edge_weight_forward.data_ptr<float>(),
+      (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+      (const uint32_t *)c.column_offset.data_ptr<int32_t>(),
+      c.src_range[0], c.src_range[1], c.dst_range[0], c.dst_range[1],
+      c.edge_size, c.dst_n(), (uint32_t)x.size(1), 1);
+}
+
+inline void csr_backward(nts_stream *s, const CSC_segment_pinned &c,
+                         const NtsVar &g, NtsVar &out) {
+  assert(g.is_cuda() && g.is_contiguous() &&
+         g.size(0) == (int64_t)c.dst_n());
+  nts_gather_by_src_from_dst(
+      s, g.data_ptr<float>(), out.data_ptr<float>(),
+      c.edge_weight_backward.data_ptr<float>(),
+      (const uint32_t *)c.row_offset.data_ptr<int32_t>(),
+      (const uint32_t *)c.column_indices.data_ptr<int32_t>(),
+      c.src_range[0], c.src_range[1], c.dst_range[0], c.dst_range[1],
+      c.edge_size, c.src_n(), (uint32_t)g.size(1), 1);
+}
+
+}  // namespace detail
+
+/* Single-GPU fused aggregation op (ForwardSingleGPUfuseOp surface,
+ * core/ntsSingleGPUFusedGraphOp.hpp:48-71): the whole graph is one chunk. */
+class ForwardSingleGPUfuseOp : public ntsGraphOp {
+ public:
+  ForwardSingleGPUfuseOp(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {}
+  NtsVar forward(NtsVar &f_input) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar y = torch::zeros({(int64_t)c.dst_n(), f_input.size(1)},
+                            f_input.options());
+    detail::csc_forward(partitioned_graph_->stream, c, f_input, y);
+    return y;
+  }
+  NtsVar backward(NtsVar &output_grad) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar gx = torch::zeros({(int64_t)c.src_n(), output_grad.size(1)},
+                             output_grad.options());
+    detail::csr_backward(partitioned_graph_->stream, c, output_grad, gx);
+    return gx;
+  }
+};
+
+}  // namespace op
+
+/* Tape-based context (NtsContext surface, core/ntsContext.hpp:108-359):
+ * graph ops bypass libtorch autograd; NN segments use it.  self_backward
+ * walks the tape mixing torch::autograd::grad with ntsGraphOp::backward. */
+class NtsContext {
+  enum Kind { GRAPHOP, NNOP };
+  struct Entry {
+    Kind kind;
+    NtsVar input, output;
+    op::ntsGraphOp *op = nullptr;
+  };
+  std::vector<Entry> tape_;
+
+ public:
+  ~NtsContext() { reset(); }
+
+  template <typename GOPT>
+  NtsVar runGraphOp(PartitionedGraph *pg, VertexSubset *active,
+                    NtsVar &f_input) {
+    auto *op = new GOPT(pg, active);
+    NtsVar out = op->forward(f_input);
+    tape_.push_back({GRAPHOP, f_input, out, op});
+    return out;
+  }
+
+  /* vertexForward: an NN function applied under torch autograd
+   * (ntsContext.hpp:198-226 semantics, consecutive NN ops chained). */
+  NtsVar runVertexForward(std::function<NtsVar(NtsVar &)> f, NtsVar &input) {
+    NtsVar in_leaf = input.detach().requires_grad_(true);
+    NtsVar out = f(in_leaf);
+    tape_.push_back({NNOP, in_leaf, out, nullptr});
+    return out;
+  }
+
+  /* Walk the tape backward from `loss` (ntsContext.hpp:276-359). Returns the
+   * gradient w.r.t. the first entry's input. */
+  NtsVar self_backward(NtsVar &loss) {
+    NtsVar grad;  // grad w.r.t. current entry's OUTPUT
+    for (auto it = tape_.rbegin(); it != tape_.rend(); ++it) {
+      if (it->kind == NNOP) {
+        /* accumulate into every reachable leaf (layer weights) AND the
+         * segment input, like torch::backward at ntsContext.hpp:283-301 */
+        if (!grad.defined()) {
+          loss.backward();
+        } else {
+          torch::autograd::backward({it->output}, {grad});
+        }
+        grad = it->input.grad();
+      } else {
+        assert(grad.defined());
+        grad = grad.contiguous();
+        grad = it->op->backward(grad);
+      }
+    }
+    reset();
+    return grad;
+  }
+
+  void reset() {
+    for (auto &e : tape_)
+      if (e.op) delete e.op;  /* context owns graph ops, ntsContext.hpp:266-275 */
+    tape_.clear();
+  }
+  size_t tape_size() const { return tape_.size(); }
+};
+
+}  // namespace nts

@@ -161,7 +161,7 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
     const uint32_t fbase = slab * slabf;
 
     float acc[ELEM * NSTR] = {};
-    if (ELEM == 1) {
+    if constexpr (ELEM == 1) {
       /* strided dword path: element j at fbase + glane + j*G */
       bool any = fbase + glane < f;
       if (any) {
@@ -192,7 +192,7 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
           const uint64_t src = nbr[e] - nbr_start;
           const float w = WITH_W ? ew[e] : 1.0f;
           const float *p = in + src * f + off;
-          if (ELEM == 4) {
+          if constexpr (ELEM == 4) {
             const float4 x = *reinterpret_cast<const float4 *>(p);
             acc[0] = fmaf(w, x.x, acc[0]);
             acc[1] = fmaf(w, x.y, acc[1]);
@@ -215,14 +215,16 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
       if (nv > 0) {
         float *o = out + (uint64_t)v * f + off;
         if (!shared_v) {
-          if (nv == 4) {
-            float4 y = *reinterpret_cast<float4 *>(o);
-            y.x += acc[0]; y.y += acc[1]; y.z += acc[2]; y.w += acc[3];
-            *reinterpret_cast<float4 *>(o) = y;
-          } else if (nv == 2 && ELEM == 2) {
-            float2 y = *reinterpret_cast<float2 *>(o);
-            y.x += acc[0]; y.y += acc[1];
-            *reinterpret_cast<float2 *>(o) = y;
+          if (nv == ELEM) {
+            if constexpr (ELEM == 4) {
+              float4 y = *reinterpret_cast<float4 *>(o);
+              y.x += acc[0]; y.y += acc[1]; y.z += acc[2]; y.w += acc[3];
+              *reinterpret_cast<float4 *>(o) = y;
+            } else {
+              float2 y = *reinterpret_cast<float2 *>(o);
+              y.x += acc[0]; y.y += acc[1];
+              *reinterpret_cast<float2 *>(o) = y;
+            }
           } else {
             for (int j = 0; j < nv; ++j) o[j] += acc[j];
           }
@@ -359,6 +361,26 @@ __global__ void k_copy(float *__restrict__ dst, const float *__restrict__ src,
     dst[i] = src[i];
 }
 
+/* NTS_DEBUG_SYNC=1: synchronize + error-check after every launch (debug). */
+bool debug_sync_on() {
+  static int v = -1;
+  if (v < 0) {
+    const char *e = getenv("NTS_DEBUG_SYNC");
+    v = (e && *e && *e != '0') ? 1 : 0;
+  }
+  return v != 0;
+}
+
+void dbg_sync(nts_stream *s, const char *what) {
+  if (!debug_sync_on()) return;
+  hipError_t e = hipStreamSynchronize(s->stream);
+  if (e == hipSuccess) e = hipGetLastError();
+  if (e != hipSuccess) {
+    fprintf(stderr, "nts_hip DEBUG after %s: %s\n", what, hipGetErrorString(e));
+    abort();
+  }
+}
+
 uint32_t grid_for(uint64_t threads) {
   uint64_t b = (threads + NTS_BLOCK - 1) / NTS_BLOCK;
   if (b < 1) b = 1;
@@ -382,6 +404,7 @@ ItemsBuf &get_items(nts_stream *s, const uint32_t *offset, uint32_t batch,
                        s->stream, offset, batch, NTS_SPLIT, ib.items,
                        ib.counter);
   }
+  dbg_sync(s, "k_build_items");
   auto res = s->items_cache.emplace(key, ib);
   return res.first->second;
 }
@@ -421,6 +444,7 @@ void launch_gather(nts_stream *s, const float *in, float *out, const float *ew,
     else NTS_LAUNCH(1, false);
   }
 #undef NTS_LAUNCH
+  dbg_sync(s, "k_gather_spmm");
 }
 }  // namespace
 
