@@ -1,0 +1,121 @@
+"""Single-GPU GAT layer over the edge-valued kernel path (BASELINE config #5).
+
+Mirrors the decomposed op chain the reference runs for GAT
+(/root/reference/toolkits/GAT_GPU_DIST.hpp:191-215 via the DistGPU*Op classes,
+core/ntsDistGPUGraphOp.hpp:145-361), specialized to one partition (the
+single-GPU case: every vertex is its own master, mirror_index = identity):
+
+  per layer, with projected features H = X·W (torch, plumbing) and attention
+  vectors a_src/a_dst:
+    1. s_src[v] = H[v]·a_src, s_dst[v] = H[v]·a_dst          (torch mv)
+    2. m_src = scatter_src_mirror_to_msg(s_src)   per-edge scalar  (E×1)
+       m_dst = scatter_dst_to_msg(s_dst)                          (E×1)
+    3. e = leaky_relu(m_src + m_dst)              (torch, elementwise)
+    4. s = edge_softmax_forward(e)                per-dst, cached
+    5. y = gather_by_dst_from_src(H, weight=s)    the SpMM hot kernel with
+                                                  per-edge attention weights
+  backward: edge_softmax_backward, scatter of grads to edges, CSR gather
+  with the CSC->CSR-permuted attention weights, and the msg->vertex reducers.
+
+The per-edge attention value is a SCALAR (f=1), matching the exercised
+reference config: E×f edge tensors never materialize.
+"""
+import numpy as np
+import torch
+
+from . import shim
+from .graph import Chunk
+from .ops import DeviceChunk, _u32_cuda
+
+
+class GATLayer:
+    """Attention-weighted aggregation on one chunk (whole graph on 1 GPU)."""
+
+    def __init__(self, ch: Chunk, v: int, device):
+        self.np_ch = ch
+        self.ch = DeviceChunk(ch, device)
+        self.stream = shim.Stream.wrap_torch_current()
+        self.mirror_index = _u32_cuda(np.arange(v, dtype=np.uint32), device)
+        self.device = device
+        self.E = ch.edge_size
+        deg = np.diff(ch.column_offset.astype(np.int64))
+        dst_of_edge = np.repeat(np.arange(ch.dst_n, dtype=np.int64), deg)
+        src_of_edge = ch.row_indices.astype(np.int64) - ch.src_s
+        self.dst_of_edge = torch.from_numpy(dst_of_edge).to(device)
+        self.src_of_edge = torch.from_numpy(src_of_edge).to(device)
+        # CSC -> CSR permutation: sort CSC edges by src (stable), which is
+        # exactly the CSR construction order (graph.build_chunks sorts the
+        # same underlying edge sequence stably by dst then by src).
+        perm = np.argsort(src_of_edge, kind="stable")
+        self.csr_from_csc = torch.from_numpy(perm).to(device)
+
+    def forward(self, h: torch.Tensor, s_src: torch.Tensor,
+                s_dst: torch.Tensor, negative_slope: float = 0.2):
+        """h: [V,f] projected features; s_src/s_dst: [V] attention scalars."""
+        ch, st, E = self.ch, self.stream, self.E
+        dev = h.device
+        h = h.contiguous()
+        m_src = torch.empty(E, 1, device=dev)
+        m_dst = torch.empty(E, 1, device=dev)
+        st.scatter_src_mirror_to_msg(m_src.data_ptr(),
+                                     s_src.contiguous().data_ptr(),
+                                     ch.row_indices.data_ptr(),
+                                     ch.column_offset.data_ptr(),
+                                     self.mirror_index.data_ptr(),
+                                     ch.dst_n, 1)
+        st.scatter_dst_to_msg(m_dst.data_ptr(), s_dst.contiguous().data_ptr(),
+                              ch.row_indices.data_ptr(),
+                              ch.column_offset.data_ptr(), ch.dst_n, 1)
+        m_sum = m_src + m_dst
+        e = torch.nn.functional.leaky_relu(m_sum, negative_slope)
+        s = torch.empty(E, 1, device=dev)
+        cached = torch.empty(E, 1, device=dev)
+        st.edge_softmax_forward(s.data_ptr(), e.contiguous().data_ptr(),
+                                cached.data_ptr(), ch.row_indices.data_ptr(),
+                                ch.column_offset.data_ptr(), ch.dst_n, 1)
+        y = torch.zeros(ch.dst_n, h.shape[1], device=dev)
+        st.gather_by_dst_from_src(h.data_ptr(), y.data_ptr(), s.data_ptr(),
+                                  ch.row_indices.data_ptr(),
+                                  ch.column_offset.data_ptr(),
+                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
+                                  E, ch.dst_n, h.shape[1], with_weight=True)
+        saved = {"h": h, "s": s, "cached": cached, "m_sum": m_sum}
+        return y, saved
+
+    def backward(self, grad_y: torch.Tensor, saved,
+                 negative_slope: float = 0.2):
+        """Returns (grad_h from the aggregation, grad_s_src[V], grad_s_dst[V])."""
+        ch, st, E = self.ch, self.stream, self.E
+        dev = grad_y.device
+        f = grad_y.shape[1]
+        grad_y = grad_y.contiguous()
+        # CSR gather with attention weights permuted into CSR edge order
+        s_bwd = saved["s"][self.csr_from_csc].contiguous()
+        grad_h = torch.zeros(ch.src_n, f, device=dev)
+        st.gather_by_src_from_dst(grad_y.data_ptr(), grad_h.data_ptr(),
+                                  s_bwd.data_ptr(),
+                                  ch.row_offset.data_ptr(),
+                                  ch.column_indices.data_ptr(),
+                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
+                                  E, ch.src_n, f, with_weight=True)
+        # d y / d s[e] = grad_y[dst(e)] . h[src(e)]
+        gs = (grad_y[self.dst_of_edge] *
+              saved["h"][self.src_of_edge]).sum(1, keepdim=True)
+        ge = torch.empty(E, 1, device=dev)
+        st.edge_softmax_backward(ge.data_ptr(), gs.contiguous().data_ptr(),
+                                 saved["cached"].data_ptr(),
+                                 ch.row_indices.data_ptr(),
+                                 ch.column_offset.data_ptr(), ch.dst_n, 1)
+        ge = ge * torch.where(saved["m_sum"] > 0, 1.0, negative_slope)
+        ge = ge.contiguous()
+        g_src = torch.zeros(ch.src_n, 1, device=dev)
+        st.gather_msg_to_src_mirror(g_src.data_ptr(), ge.data_ptr(),
+                                    ch.row_indices.data_ptr(),
+                                    ch.column_offset.data_ptr(),
+                                    self.mirror_index.data_ptr(),
+                                    ch.dst_n, 1)
+        g_dst = torch.zeros(ch.dst_n, 1, device=dev)
+        st.gather_msg_to_dst(g_dst.data_ptr(), ge.data_ptr(),
+                             ch.row_indices.data_ptr(),
+                             ch.column_offset.data_ptr(), ch.dst_n, 1)
+        return grad_h, g_src[:, 0], g_dst[:, 0]

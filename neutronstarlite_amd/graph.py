@@ -153,10 +153,16 @@ def build_chunks(edges: np.ndarray, weights: np.ndarray, offs: np.ndarray,
         src_s, src_e = int(offs[k]), int(offs[k + 1])
         src_n = src_e - src_s
         col_off, rows, wf = _csc(s, d, w, dst_s, dst_n)
-        # CSR: sort by src
-        order = np.argsort(s.astype(np.int64), kind="stable")
+        # CSR built as a stable-by-src permutation OF THE CSC ORDER, so that
+        # csr_edge[i] == csc_edge[argsort_stable(src_of_csc)[i]] — the
+        # edge-valued (GAT) path relies on this to carry per-edge attention
+        # weights from CSC to CSR order (gat.py csr_from_csc).
+        d_csc = np.repeat(np.arange(dst_n, dtype=np.int64),
+                          np.diff(col_off.astype(np.int64))) + dst_s
+        order = np.argsort(rows.astype(np.int64), kind="stable")
         row_off = np.zeros(src_n + 1, dtype=np.uint32)
-        counts = np.bincount((s[order] - src_s).astype(np.int64), minlength=src_n)
+        counts = np.bincount((rows[order] - src_s).astype(np.int64),
+                             minlength=src_n)
         row_off[1:] = np.cumsum(counts).astype(np.uint32)
         chunks.append(Chunk(
             src_s=src_s, src_e=src_e, dst_s=dst_s, dst_e=dst_e,
@@ -164,8 +170,8 @@ def build_chunks(edges: np.ndarray, weights: np.ndarray, offs: np.ndarray,
             row_indices=rows.astype(np.uint32),
             edge_weight_forward=wf.astype(np.float32),
             row_offset=row_off.astype(np.uint32),
-            column_indices=d[order].astype(np.uint32),
-            edge_weight_backward=w[order].astype(np.float32),
+            column_indices=d_csc[order].astype(np.uint32),
+            edge_weight_backward=wf[order].astype(np.float32),
         ))
     return chunks
 

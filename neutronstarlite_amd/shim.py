@@ -3,8 +3,10 @@
 FAILS LOUDLY if the HIP extension is missing: there is no CPU fallback in the
 product path — the oracle under oracle/ is test infrastructure only.
 """
+import atexit
 import ctypes
 import os
+import weakref
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 _SO = os.path.join(_DIR, "libnts_hip.so")
@@ -91,6 +93,21 @@ def lib():
     return l
 
 
+_live_streams = weakref.WeakSet()
+
+
+@atexit.register
+def _destroy_streams_at_exit():
+    # Destroy stream objects (draining events, freeing item caches) while the
+    # HIP runtime is still alive; leaking them into runtime teardown can
+    # abort the process at exit.
+    for s in list(_live_streams):
+        try:
+            s.destroy()
+        except Exception:
+            pass
+
+
 class Stream:
     """Thin RAII wrapper over nts_stream (replaces Cuda_Stream,
     /root/reference/cuda/ntsCUDA.hpp:97-217)."""
@@ -102,6 +119,7 @@ class Stream:
         else:
             self.h = l.nts_stream_create()
         self._lib = l
+        _live_streams.add(self)
 
     @classmethod
     def wrap_torch_current(cls):
