@@ -116,6 +116,10 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--feat", type=int, default=602)
+    ap.add_argument("--model", default="gcn", choices=["gcn", "gat"],
+                    help="gcn = fused norm-degree aggregation (configs #2-4);"
+                         " gat = attention-weighted layer with edge softmax "
+                         "(config #5, single GPU, --feat 128)")
     ap.add_argument("--graph", default="reddit",
                     choices=["reddit", "rmat26", "small"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -204,10 +208,27 @@ def main():
     gy = torch.from_numpy(
         rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
 
-    def step():
-        y = ring_forward(rg, x, engine)
-        gx = ring_backward(rg, gy, engine)
-        return y, gx
+    if args.model == "gat":
+        # BASELINE config #5: 1-GPU attention-weighted layer; the hot SpMM
+        # runs with per-edge softmax weights; edge kernels (a14) feed it.
+        assert not distributed, "GAT bench is the single-GPU config (#5)"
+        from neutronstarlite_amd.gat import GATLayer
+        layer = GATLayer(chunks[0], v, dev)
+        a_src = torch.from_numpy(
+            rng.uniform(-1, 1, size=f).astype(np.float32)).to(dev)
+        a_dst = torch.from_numpy(
+            rng.uniform(-1, 1, size=f).astype(np.float32)).to(dev)
+        # reuse the engine's timing stream for the roofline numbers
+        layer.stream = engine.stream
+
+        def step():
+            y, saved = layer.forward(x, x @ a_src, x @ a_dst)
+            return layer.backward(gy, saved)
+    else:
+        def step():
+            y = ring_forward(rg, x, engine)
+            gx = ring_backward(rg, gy, engine)
+            return y, gx
 
     log(f"rank {rank}/{n}: warmup {args.warmup}")
     for _ in range(args.warmup):
@@ -271,11 +292,16 @@ def main():
             "dtype": "f32",
             "data": "synthetic",
             "config": {
-                "workload": (f"{args.graph}: RMAT V={v} E={e_total} "
-                             f"feat={f} GCN-layer aggregation fwd+bwd "
-                             "(BASELINE config #2)" if args.graph == "reddit"
-                             else f"{args.graph}: RMAT V={v} E={e_total} feat={f}"),
-                "V": v, "E": e_total, "feat": f,
+                "workload": (
+                    f"{args.graph}: RMAT V={v} E={e_total} feat={f} "
+                    + ("GAT layer (edge softmax + attention-weighted "
+                       "aggregation) fwd+bwd (BASELINE config #5)"
+                       if args.model == "gat" else
+                       "GCN-layer aggregation fwd+bwd"
+                       + (" (BASELINE config #2)" if args.graph == "reddit"
+                          else ""))),
+                "model": args.model, "V": v, "E": e_total, "feat": f,
+                "relabel": args.relabel,
                 "parallelism": f"graph-partitioned dp{n}",
                 "edges_per_step": 2 * e_total,
             },

@@ -173,6 +173,16 @@ void nts_edge_softmax_backward(nts_stream *s, float *msg_input_grad,
     const nts_vid *row_indices, const nts_vid *column_offset,
     nts_vid batch_size, nts_vid feature_size);
 
+/* Per-edge dot product (additive entry point): for each edge e of dst d,
+ *   out[e] = dot(dst_rows[d,:], src_rows[row_indices[e]-src_start,:]).
+ * The GAT backward needs d y/d s[e] = grad_y[dst(e)] . h[src(e)]
+ * (ntsDistGPUGraphOp.hpp's chain materializes E x f edge tensors for this;
+ * here the dot is fused so only E scalars ever exist). */
+void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
+    const float *src_rows, const nts_vid *row_indices,
+    const nts_vid *column_offset, nts_vid src_start, nts_vid batch_size,
+    nts_vid feature_size);
+
 /* Replaces Cuda_Stream::Scatter_Grad_Back_To_Message (ntsCUDA.hpp:193-198;
  * kernel scatter_grad_back_to_messaage, ntsCUDAFuseKernel.cuh:492-506):
  *   message_grad[e,:] += input_grad[d,:] for each edge e of dst d. */

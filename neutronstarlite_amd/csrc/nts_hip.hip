@@ -317,6 +317,31 @@ __global__ void k_edge_op(float *__restrict__ message,
   }
 }
 
+/* per-edge dot: out[e] = dot(dst_rows[d], src_rows[row_indices[e]-src_s]),
+ * one wavefront per destination, lanes strided over the feature dim. */
+__global__ void k_edge_dot(float *__restrict__ out,
+                           const float *__restrict__ dst_rows,
+                           const float *__restrict__ src_rows,
+                           const uint32_t *__restrict__ row_indices,
+                           const uint32_t *__restrict__ column_offset,
+                           uint32_t src_start, uint32_t batch, uint32_t f) {
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t d = wave; d < batch; d += n_waves) {
+    const uint32_t e0 = column_offset[d], e1 = column_offset[d + 1];
+    const float *a = dst_rows + (uint64_t)d * f;
+    for (uint32_t e = e0; e < e1; ++e) {
+      const float *b = src_rows + (uint64_t)(row_indices[e] - src_start) * f;
+      float sum = 0.f;
+      for (uint32_t j = lane; j < f; j += 64) sum += a[j] * b[j];
+#pragma unroll
+      for (int w = 32; w >= 1; w >>= 1) sum += __shfl_xor(sum, w, 64);
+      if (lane == 0) out[e] = sum;
+    }
+  }
+}
+
 /* per-dst softmax over incident edge values, per feature slot.
  * Restates edge_softmax_forward_block (ntsCUDADistKernel.cuh:166-213):
  * out[e] = exp(in[e]) / sum_{e' in dst} exp(in[e']), no max subtraction
@@ -694,6 +719,19 @@ void nts_gather_msg_to_dst(nts_stream *s, float *dst_feature,
                            nts_vid feature_size) {
   launch_edge_op(s, E_GATHER_DST, const_cast<float *>(message), dst_feature,
                  row_indices, column_offset, nullptr, batch_size, feature_size);
+}
+
+void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
+                  const float *src_rows, const nts_vid *row_indices,
+                  const nts_vid *column_offset, nts_vid src_start,
+                  nts_vid batch_size, nts_vid feature_size) {
+  if (!batch_size || !feature_size) return;
+  Tic t(s, NTS_KTAG_EDGE);
+  hipLaunchKernelGGL(k_edge_dot, dim3(grid_for((uint64_t)batch_size * 64)),
+                     dim3(NTS_BLOCK), 0, s->stream, out, dst_rows, src_rows,
+                     row_indices, column_offset, src_start, batch_size,
+                     feature_size);
+  dbg_sync(s, "k_edge_dot");
 }
 
 void nts_scatter_grad_back_to_message(nts_stream *s, const float *input_grad,

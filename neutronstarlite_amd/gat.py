@@ -39,15 +39,26 @@ class GATLayer:
         self.device = device
         self.E = ch.edge_size
         deg = np.diff(ch.column_offset.astype(np.int64))
-        dst_of_edge = np.repeat(np.arange(ch.dst_n, dtype=np.int64), deg)
-        src_of_edge = ch.row_indices.astype(np.int64) - ch.src_s
-        self.dst_of_edge = torch.from_numpy(dst_of_edge).to(device)
-        self.src_of_edge = torch.from_numpy(src_of_edge).to(device)
+        self._dst_of_edge_np = np.repeat(np.arange(ch.dst_n, dtype=np.int64), deg)
+        self._src_of_edge_np = ch.row_indices.astype(np.int64) - ch.src_s
         # CSC -> CSR permutation: sort CSC edges by src (stable), which is
-        # exactly the CSR construction order (graph.build_chunks sorts the
-        # same underlying edge sequence stably by dst then by src).
-        perm = np.argsort(src_of_edge, kind="stable")
+        # exactly the CSR construction order (graph.build_chunks builds CSR
+        # as the stable-by-src permutation of the CSC order).
+        perm = np.argsort(self._src_of_edge_np, kind="stable")
         self.csr_from_csc = torch.from_numpy(perm).to(device)
+        self._doe = self._soe = None
+
+    @property
+    def dst_of_edge(self):
+        if self._doe is None:
+            self._doe = torch.from_numpy(self._dst_of_edge_np).to(self.device)
+        return self._doe
+
+    @property
+    def src_of_edge(self):
+        if self._soe is None:
+            self._soe = torch.from_numpy(self._src_of_edge_np).to(self.device)
+        return self._soe
 
     def forward(self, h: torch.Tensor, s_src: torch.Tensor,
                 s_dst: torch.Tensor, negative_slope: float = 0.2):
@@ -98,9 +109,12 @@ class GATLayer:
                                   ch.column_indices.data_ptr(),
                                   ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
                                   E, ch.src_n, f, with_weight=True)
-        # d y / d s[e] = grad_y[dst(e)] . h[src(e)]
-        gs = (grad_y[self.dst_of_edge] *
-              saved["h"][self.src_of_edge]).sum(1, keepdim=True)
+        # d y / d s[e] = grad_y[dst(e)] . h[src(e)] — fused edge-dot kernel
+        # (E scalars; never materializes E x f edge tensors)
+        gs = torch.empty(E, 1, device=dev)
+        st.edge_dot(gs.data_ptr(), grad_y.data_ptr(),
+                    saved["h"].data_ptr(), ch.row_indices.data_ptr(),
+                    ch.column_offset.data_ptr(), ch.src_s, ch.dst_n, f)
         ge = torch.empty(E, 1, device=dev)
         st.edge_softmax_backward(ge.data_ptr(), gs.contiguous().data_ptr(),
                                  saved["cached"].data_ptr(),

@@ -84,6 +84,7 @@ def lib():
     l.nts_scatter_dst_to_msg.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
     l.nts_gather_msg_to_dst.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
     l.nts_scatter_grad_back_to_message.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
+    l.nts_edge_dot.argtypes = [_vp] + [_vp] * 4 + [_u32] * 3
     l.nts_edge_softmax_forward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_device_count.restype = _i32
@@ -208,6 +209,12 @@ class Stream:
         self._lib.nts_gather_msg_to_dst(self.h, _vp(dst_feat), _vp(msg),
                                         _vp(row_indices), _vp(column_offset),
                                         batch, f)
+
+    def edge_dot(self, out, dst_rows, src_rows, row_indices, column_offset,
+                 src_start, batch, f):
+        self._lib.nts_edge_dot(self.h, _vp(out), _vp(dst_rows), _vp(src_rows),
+                               _vp(row_indices), _vp(column_offset),
+                               src_start, batch, f)
 
     def scatter_grad_back_to_message(self, input_grad, msg_grad, row_indices,
                                      column_offset, batch, f):
