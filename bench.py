@@ -135,6 +135,14 @@ def main():
                          "rocprofv3 --pmc pass (see profiles/)")
     args = ap.parse_args()
 
+    if args.traffic_bytes_per_launch is None:
+        # measured HBM bytes/launch from the separate rocprofv3 --pmc passes
+        # committed under profiles/ (FETCH corrected x2 per the gfx950
+        # wide-read calibration + WRITE), keyed by exact profiled config
+        args.traffic_bytes_per_launch = {
+            ("reddit", 602, "none", "gcn"): 2.62e11,  # profiles/round1
+        }.get((args.graph, args.feat, args.relabel, args.model))
+
     import torch
     import torch.distributed as dist
     from neutronstarlite_amd import graph as G, shim
