@@ -90,7 +90,9 @@ struct PartitionedGraph {
   std::vector<VertexId> partition_offset;  // [P+1]
   nts_stream *stream = nullptr;            // compute stream (C-ABI)
 
-  PartitionedGraph() { stream = nts_stream_create(); }
+  /* Wrap the HIP null stream (= torch's default stream): libtorch tensor
+   * fills/copies and our kernels must share one stream order. */
+  PartitionedGraph() { stream = nts_stream_wrap(nullptr); }
   ~PartitionedGraph() {
     if (stream) nts_stream_destroy(stream);
   }

@@ -115,10 +115,13 @@ class Stream:
 
     def __init__(self, wrap_hip_stream=None):
         l = lib()
-        if wrap_hip_stream:
-            self.h = l.nts_stream_wrap(_vp(wrap_hip_stream))
-        else:
+        if wrap_hip_stream is None:
             self.h = l.nts_stream_create()
+        else:
+            # NOTE: torch's default stream handle is 0 (the HIP null stream);
+            # it MUST be wrapped, not replaced — launching on a private
+            # non-blocking stream would race torch's fills/copies.
+            self.h = l.nts_stream_wrap(_vp(wrap_hip_stream))
         self._lib = l
         _live_streams.add(self)
 

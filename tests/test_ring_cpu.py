@@ -15,7 +15,6 @@ from neutronstarlite_amd import graph as G
 from neutronstarlite_amd.ring import RingGraph, ring_backward, ring_forward
 
 V, E, F, SEED = 777, 9000, 11, 7
-WORLD = 2
 
 
 class OracleEngine:
