@@ -33,24 +33,24 @@ class OracleEngine:
                             ch.dst_s, ch.src_n, out.shape[1], out=out.numpy())
 
 
-def _data():
+def _data(world):
     edges = G.rmat_edges(V, E, seed=SEED)
     outd, ind = G.degrees(edges, V)
     w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
-    offs = G.partition_offsets(edges, V, WORLD)
+    offs = G.partition_offsets(edges, V, world)
     rng = np.random.default_rng(42)
     x = rng.uniform(-1, 1, size=(V, F)).astype(np.float32)
     g = rng.uniform(-1, 1, size=(V, F)).astype(np.float32)
     return edges, w, offs, x, g
 
 
-def _worker(rank, tmpdir, q):
+def _worker(rank, world, tmpdir, q):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         dist.init_process_group(
             "gloo", init_method=f"file://{tmpdir}/pg", rank=rank,
-            world_size=WORLD)
-        edges, w, offs, x, g = _data()
+            world_size=world)
+        edges, w, offs, x, g = _data(world)
         chunks = G.build_chunks(edges, w, offs, rank)
         rg = RingGraph(offs, rank, chunks, torch.device("cpu"))
         lo, hi = int(offs[rank]), int(offs[rank + 1])
@@ -64,16 +64,17 @@ def _worker(rank, tmpdir, q):
         raise
 
 
-@pytest.mark.timeout(180)
-def test_ring_matches_whole_graph(tmp_path):
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("world", [2, 3])
+def test_ring_matches_whole_graph(tmp_path, world):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_worker, args=(r, str(tmp_path), q))
-             for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker, args=(r, world, str(tmp_path), q))
+             for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(WORLD):
+    for _ in range(world):
         rank, y, gx = q.get()
         assert not (isinstance(y, str) and y == "error"), f"rank {rank}: {gx}"
         results[rank] = (y, gx)
@@ -81,13 +82,13 @@ def test_ring_matches_whole_graph(tmp_path):
         p.join(60)
         assert p.exitcode == 0
 
-    edges, w, offs, x, g = _data()
+    edges, w, offs, x, g = _data(world)
     ch = G.build_chunks(edges, w, np.array([0, V], dtype=np.uint32), 0)[0]
     y_ref = oracle.csc_forward(ch.column_offset, ch.row_indices,
                                ch.edge_weight_forward, x, 0, V, F)
     gx_ref = oracle.csr_backward(ch.row_offset, ch.column_indices,
                                  ch.edge_weight_backward, g, 0, V, F)
-    y_all = np.concatenate([results[r][0] for r in range(WORLD)])
-    gx_all = np.concatenate([results[r][1] for r in range(WORLD)])
+    y_all = np.concatenate([results[r][0] for r in range(world)])
+    gx_all = np.concatenate([results[r][1] for r in range(world)])
     assert np.allclose(y_all, y_ref, rtol=1e-4, atol=1e-5)
     assert np.allclose(gx_all, gx_ref, rtol=1e-4, atol=1e-5)
