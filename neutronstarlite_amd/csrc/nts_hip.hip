@@ -44,9 +44,26 @@
   } while (0)
 
 namespace {
-constexpr uint32_t NTS_SPLIT = 256;   /* max edges per work item */
 constexpr int NTS_BLOCK = 256;        /* 4 waves */
-constexpr int NTS_MAX_BLOCKS = 2048;  /* 8 blocks/CU on 256 CUs */
+
+/* Tunables (defaults chosen by measurement on gfx950; env-overridable for
+ * sweeps: NTS_SPLIT = max edges per work item, NTS_MAX_BLOCKS = grid cap). */
+uint32_t env_u32(const char *name, uint32_t dflt) {
+  const char *e = getenv(name);
+  if (!e || !*e) return dflt;
+  long v = atol(e);
+  return v > 0 ? (uint32_t)v : dflt;
+}
+uint32_t nts_split() {
+  static uint32_t v = env_u32("NTS_SPLIT", 256);
+  return v;
+}
+uint32_t nts_max_blocks() {
+  static uint32_t v = env_u32("NTS_MAX_BLOCKS", 2048);
+  return v;
+}
+#define NTS_SPLIT nts_split()
+#define NTS_MAX_BLOCKS nts_max_blocks()
 
 struct ItemsBuf {
   uint4 *items = nullptr;
