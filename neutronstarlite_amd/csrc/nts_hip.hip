@@ -182,7 +182,26 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
       /* strided dword path: element j at fbase + glane + j*G */
       bool any = fbase + glane < f;
       if (any) {
-        for (uint32_t e = e0; e < e0 + cnt; ++e) {
+        /* 4-edge unroll: 4 independent row loads in flight per wave
+         * (one dependent load per iteration leaves HBM latency exposed) */
+        uint32_t e = e0;
+        const uint32_t e_end = e0 + cnt;
+        for (; e + 4 <= e_end; e += 4) {
+          const float *p[4];
+          float wv[4];
+#pragma unroll
+          for (int k = 0; k < 4; ++k) {
+            p[k] = in + (uint64_t)(nbr[e + k] - nbr_start) * f + fbase + glane;
+            wv[k] = WITH_W ? ew[e + k] : 1.0f;
+          }
+#pragma unroll
+          for (int k = 0; k < 4; ++k)
+#pragma unroll
+            for (int j = 0; j < NSTR; ++j)
+              if (fbase + glane + j * G < f)
+                acc[j] = fmaf(wv[k], p[k][j * G], acc[j]);
+        }
+        for (; e < e_end; ++e) {
           const uint64_t src = nbr[e] - nbr_start;
           const float w = WITH_W ? ew[e] : 1.0f;
           const float *p = in + src * f + fbase + glane;
@@ -205,7 +224,42 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
       const uint32_t off = fbase + glane * ELEM;
       const int nv = (off + ELEM <= f) ? ELEM : (off < f ? (int)(f - off) : 0);
       if (nv == ELEM) {
-        for (uint32_t e = e0; e < e0 + cnt; ++e) {
+        /* 4-edge unroll: 4 independent vector row loads in flight */
+        uint32_t e = e0;
+        const uint32_t e_end = e0 + cnt;
+        for (; e + 4 <= e_end; e += 4) {
+          const float *p[4];
+          float wv[4];
+#pragma unroll
+          for (int k = 0; k < 4; ++k) {
+            p[k] = in + (uint64_t)(nbr[e + k] - nbr_start) * f + off;
+            wv[k] = WITH_W ? ew[e + k] : 1.0f;
+          }
+          if constexpr (ELEM == 4) {
+            float4 x[4];
+#pragma unroll
+            for (int k = 0; k < 4; ++k)
+              x[k] = *reinterpret_cast<const float4 *>(p[k]);
+#pragma unroll
+            for (int k = 0; k < 4; ++k) {
+              acc[0] = fmaf(wv[k], x[k].x, acc[0]);
+              acc[1] = fmaf(wv[k], x[k].y, acc[1]);
+              acc[2] = fmaf(wv[k], x[k].z, acc[2]);
+              acc[3] = fmaf(wv[k], x[k].w, acc[3]);
+            }
+          } else {
+            float2 x[4];
+#pragma unroll
+            for (int k = 0; k < 4; ++k)
+              x[k] = *reinterpret_cast<const float2 *>(p[k]);
+#pragma unroll
+            for (int k = 0; k < 4; ++k) {
+              acc[0] = fmaf(wv[k], x[k].x, acc[0]);
+              acc[1] = fmaf(wv[k], x[k].y, acc[1]);
+            }
+          }
+        }
+        for (; e < e_end; ++e) {
           const uint64_t src = nbr[e] - nbr_start;
           const float w = WITH_W ? ew[e] : 1.0f;
           const float *p = in + src * f + off;
