@@ -20,11 +20,15 @@
 
 using namespace nts;
 
+static void stage(const char *what) { fprintf(stderr, "[demo] %s\n", what); }
+
 int main(int argc, char **argv) {
+  stage("start");
   if (!torch::cuda::is_available()) {
     fprintf(stderr, "gcn_demo: no GPU\n");
     return 2;
   }
+  stage("cuda available");
   const int epochs = argc > 1 ? atoi(argv[1]) : 10;
   const VertexId V = 4000;
   const uint32_t E = 40000;
@@ -62,19 +66,24 @@ int main(int argc, char **argv) {
     cols[pr] = dst[e]; wb[pr] = wgt(src[e], dst[e]);
   }
 
+  stage("host graph built");
   auto chunk = CSC_segment_pinned::from_host(
       0, V, 0, V, col_off.data(), rows.data(), wf.data(), row_off.data(),
       cols.data(), wb.data(), Etot, dev);
+  stage("chunk uploaded");
   PartitionedGraph pg;
   pg.graph_chunks.push_back(&chunk);
   pg.partition_offset = {0, V};
   VertexSubset active{0, V};
 
+  stage("pg ready");
   /* parity check: one aggregation vs naive double loop on CPU */
   {
     NtsVar x = torch::rand({(int64_t)V, 8}, torch::device(dev)) * 2 - 1;
     op::ForwardSingleGPUfuseOp agg(&pg, &active);
+    stage("parity forward");
     NtsVar y = agg.forward(x).cpu();
+    stage("parity forward done");
     auto xc = x.cpu().contiguous();
     const float *xp = xc.data_ptr<float>();
     const float *yp = y.data_ptr<float>();

@@ -94,12 +94,11 @@ void nts_gather_by_src_from_dst(nts_stream *s,
     nts_vid src_start, nts_vid src_end, nts_vid dst_start, nts_vid dst_end,
     nts_vid edges, nts_vid batch_size, nts_vid feature_size, int with_weight);
 
-/* Work-item cache: the two gather entry points decompose (offset, batch)
- * into bounded per-wavefront work items on device and cache the result
- * keyed by (offset pointer, batch, feature slabs).  Chunk topology is
- * static in the reference (CSC_segment_pinned built once,
- * PartitionedGraph.hpp:324-420; CopyGraphToDevice GraphSegment.cpp:178-220),
- * so the cache is sound; call this if a topology buffer is ever rewritten. */
+/* The two gather entry points decompose (offset, batch) into bounded
+ * per-wavefront work items on device, rebuilt on every call into a
+ * stream-owned scratch buffer (caching by topology pointer would be unsound
+ * when the host frees and reallocates chunk buffers).  This entry point is
+ * retained for ABI stability; it now only synchronizes the stream. */
 void nts_items_cache_clear(nts_stream *s);
 
 /* ---- message transfer (host-bounce compatibility path) ----

@@ -13,8 +13,9 @@
  *    feature width and pointer alignment) and accumulates in REGISTERS —
  *    no LDS staging, no per-element shared-memory atomics.
  *  - Power-law load balance: columns are decomposed on device into bounded
- *    work items of <= NTS_SPLIT consecutive edges of one vertex (cached per
- *    static chunk); single-item vertices do plain read-modify-write stores,
+ *    work items of <= NTS_SPLIT consecutive edges of one vertex (rebuilt
+ *    per call into a stream scratch buffer — two launches, ~0.05% of a
+ *    gather); single-item vertices do plain read-modify-write stores,
  *    split (hub) vertices merge with device-scope fp32 atomics.
  *  - Launches are grid-stride with ~2048 blocks of 256 threads (4 waves),
  *    plenty to fill 256 CUs across 8 XCDs; occupancy (8 waves/SIMD at this
@@ -81,7 +82,7 @@ struct nts_stream {
   std::vector<std::pair<hipEvent_t, hipEvent_t>> freeev;
   double acc_ns[NTS_KTAG_COUNT] = {};
   long long acc_n[NTS_KTAG_COUNT] = {};
-  std::map<std::pair<const void *, uint32_t>, ItemsBuf> items_cache;
+  ItemsBuf items_scratch;  /* per-call work-item buffer (stream-ordered) */
 };
 
 namespace {
@@ -484,15 +485,22 @@ uint32_t grid_for(uint64_t threads) {
   return (uint32_t)b;
 }
 
+/* Build the bounded work items for (offset, batch) into the stream's
+ * scratch buffer.  Rebuilt on EVERY call: caching by topology pointer is
+ * unsound (the host may free a chunk and a later chunk's offsets can land
+ * at the same address), and the build is two launches costing ~0.05% of a
+ * gather.  All uses are ordered on the stream, so one scratch suffices. */
 ItemsBuf &get_items(nts_stream *s, const uint32_t *offset, uint32_t batch,
                     uint32_t edges) {
-  auto key = std::make_pair((const void *)offset, batch);
-  auto it = s->items_cache.find(key);
-  if (it != s->items_cache.end()) return it->second;
-  ItemsBuf ib;
-  ib.cap = (uint64_t)batch + edges / NTS_SPLIT + 1;
-  NTS_CHECK(hipMalloc(&ib.items, ib.cap * sizeof(uint4)));
-  NTS_CHECK(hipMalloc(&ib.counter, sizeof(uint32_t)));
+  ItemsBuf &ib = s->items_scratch;
+  const uint64_t need = (uint64_t)batch + edges / NTS_SPLIT + 1;
+  if (ib.cap < need) {
+    NTS_CHECK(hipStreamSynchronize(s->stream));  /* old buffer may be in use */
+    if (ib.items) NTS_CHECK(hipFree(ib.items));
+    if (!ib.counter) NTS_CHECK(hipMalloc(&ib.counter, sizeof(uint32_t)));
+    NTS_CHECK(hipMalloc(&ib.items, need * sizeof(uint4)));
+    ib.cap = need;
+  }
   NTS_CHECK(hipMemsetAsync(ib.counter, 0, sizeof(uint32_t), s->stream));
   {
     Tic t(s, NTS_KTAG_ITEMS);
@@ -501,8 +509,7 @@ ItemsBuf &get_items(nts_stream *s, const uint32_t *offset, uint32_t batch,
                        ib.counter);
   }
   dbg_sync(s, "k_build_items");
-  auto res = s->items_cache.emplace(key, ib);
-  return res.first->second;
+  return ib;
 }
 
 void launch_gather(nts_stream *s, const float *in, float *out, const float *ew,
@@ -568,10 +575,8 @@ void nts_stream_destroy(nts_stream *s) {
     hipEventDestroy(p.first);
     hipEventDestroy(p.second);
   }
-  for (auto &kv : s->items_cache) {
-    hipFree(kv.second.items);
-    hipFree(kv.second.counter);
-  }
+  if (s->items_scratch.items) hipFree(s->items_scratch.items);
+  if (s->items_scratch.counter) hipFree(s->items_scratch.counter);
   if (s->owned) hipStreamDestroy(s->stream);
   delete s;
 }
@@ -661,12 +666,8 @@ void nts_gather_by_src_from_dst(nts_stream *s, const float *input,
 }
 
 void nts_items_cache_clear(nts_stream *s) {
+  /* items are rebuilt on every call now; kept for ABI compatibility */
   NTS_CHECK(hipStreamSynchronize(s->stream));
-  for (auto &kv : s->items_cache) {
-    hipFree(kv.second.items);
-    hipFree(kv.second.counter);
-  }
-  s->items_cache.clear();
 }
 
 void nts_deserialize_to_gpu(nts_stream *s, float *gpu_buffer, const float *msg,

@@ -84,7 +84,7 @@ def lib():
     l.nts_scatter_dst_to_msg.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
     l.nts_gather_msg_to_dst.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
     l.nts_scatter_grad_back_to_message.argtypes = [_vp] + [_vp] * 4 + [_u32] * 2
-    l.nts_edge_dot.argtypes = [_vp] + [_vp] * 4 + [_u32] * 3
+    l.nts_edge_dot.argtypes = [_vp] + [_vp] * 5 + [_u32] * 3
     l.nts_edge_softmax_forward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_device_count.restype = _i32
