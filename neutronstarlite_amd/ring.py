@@ -96,18 +96,27 @@ def ring_backward(rg: RingGraph, grad_y: torch.Tensor, engine) -> torch.Tensor:
     engine.csr_backward(rg.chunks[r], grad_y, gx)  # local chunk
     if P == 1:
         return gx
-    maxn = max(rg.part_n(k) for k in range(P))
-    recv = torch.empty(maxn, f, dtype=torch.float32, device=grad_y.device)
+    # pipelined: compute step s+1's partial while step s's exchange is in
+    # flight (double-buffered receives; all ranks post exchanges in the same
+    # step order, so the grouped sends/recvs match)
+    recvs = [torch.empty(rg.owned_n, f, dtype=torch.float32,
+                         device=grad_y.device) for _ in range(2)]
+    prev = None
     for step in range(1, P):
         k = (r + step) % P          # partition whose masters we feed
         peer_src = (r - step) % P   # rank whose partial for US arrives
         partial = torch.zeros(rg.part_n(k), f, dtype=torch.float32,
                               device=grad_y.device)
         engine.csr_backward(rg.chunks[k], grad_y, partial)
-        reqs = _exchange(partial, k, recv[: rg.owned_n], peer_src)
-        for rq in reqs:
-            rq.wait()
-        gx += recv[: rg.owned_n]
+        reqs = _exchange(partial, k, recvs[step % 2], peer_src)
+        if prev is not None:
+            for rq in prev:
+                rq.wait()
+            gx += recvs[(step - 1) % 2]
+        prev = reqs
+    for rq in prev:
+        rq.wait()
+    gx += recvs[(P - 1) % 2]
     return gx
 
 
