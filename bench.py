@@ -116,10 +116,16 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--feat", type=int, default=602)
-    ap.add_argument("--model", default="gcn", choices=["gcn", "gat"],
+    ap.add_argument("--model", default="gcn",
+                    choices=["gcn", "gat", "gcn-layer"],
                     help="gcn = fused norm-degree aggregation (configs #2-4);"
                          " gat = attention-weighted layer with edge softmax "
-                         "(config #5, single GPU, --feat 128)")
+                         "(config #5, single GPU, --feat 128); gcn-layer = "
+                         "whole GCN layer with the feature-projection GEMM "
+                         "ordered BEFORE the aggregation (SURVEY 8f-2: one "
+                         "HBM pass at f_out instead of f_in when f_out<<f_in)")
+    ap.add_argument("--feat-out", type=int, default=128,
+                    help="projection width for --model gcn-layer")
     ap.add_argument("--graph", default="reddit",
                     choices=["reddit", "rmat26", "small"])
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -216,7 +222,32 @@ def main():
     gy = torch.from_numpy(
         rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
 
-    if args.model == "gat":
+    if args.model == "gcn-layer":
+        # the layer's dense projection (x·W, the reference's
+        # P[layer]->forward at NtsScheduler.hpp:737-740) ordered BEFORE the
+        # aggregation: for 602->128 the gather streams 4.7x fewer bytes.
+        # The GEMM itself is rocBLAS via torch.mm (plain library GEMM).
+        assert not distributed, "gcn-layer bench is single-GPU here"
+        f_out = args.feat_out
+        rngw = np.random.default_rng(1)
+        W = torch.from_numpy((rngw.uniform(-0.1, 0.1, size=(f, f_out)))
+                             .astype(np.float32)).to(dev)
+        y_buf = torch.zeros(hi - lo, f_out, device=dev)
+        gh_buf = torch.zeros(hi - lo, f_out, device=dev)
+        gy_s = torch.from_numpy(
+            rng.uniform(-1, 1, size=(hi - lo, f_out)).astype(np.float32)).to(dev)
+        dch = dchunks[0]
+
+        def step():
+            h = (x @ W).contiguous()
+            y_buf.zero_()
+            engine.csc_forward(dch, h, y_buf)
+            gh_buf.zero_()
+            engine.csr_backward(dch, gy_s, gh_buf)
+            gw = x.t() @ gh_buf
+            gx = gh_buf @ W.t()
+            return gw, gx
+    elif args.model == "gat":
         # BASELINE config #5: 1-GPU attention-weighted layer; the hot SpMM
         # runs with per-edge softmax weights; edge kernels (a14) feed it.
         assert not distributed, "GAT bench is the single-GPU config (#5)"
@@ -265,9 +296,11 @@ def main():
     local_edges = sum(ch.edge_size for ch in chunks)
     # algorithmic bytes per launch: per edge f floats of source
     # row + u32 index + f32 weight; per output row one read + one write.
+    # (for gcn-layer the gather runs at the projected width)
+    f_roof = args.feat_out if args.model == "gcn-layer" else f
     edges_per_launch = local_edges / max(1, len(chunks))
-    algo_bytes_launch = (edges_per_launch * (4 * f + 8)
-                         + (hi - lo) * 8 * f / max(1, len(chunks)))
+    algo_bytes_launch = (edges_per_launch * (4 * f_roof + 8)
+                         + (hi - lo) * 8 * f_roof / max(1, len(chunks)))
     avg_launch_ns = fwd_ns / max(1, fwd_launches)
     achieved = algo_bytes_launch / max(avg_launch_ns, 1e-9)  # GB/s
     roofline = {
@@ -305,6 +338,9 @@ def main():
                     + ("GAT layer (edge softmax + attention-weighted "
                        "aggregation) fwd+bwd (BASELINE config #5)"
                        if args.model == "gat" else
+                       f"GCN layer project({f}->{args.feat_out}) then "
+                       "aggregate, fwd+bwd (SURVEY 8f-2 fused-layer order)"
+                       if args.model == "gcn-layer" else
                        "GCN-layer aggregation fwd+bwd"
                        + (" (BASELINE config #2)" if args.graph == "reddit"
                           else ""))),
