@@ -289,6 +289,13 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    # per-kernel-tag breakdown over the timed region (HIP events)
+    tag_names = ["fwd", "bwd", "deser", "aggmsg", "items", "edge"]
+    log("kernel ns/step: " + "  ".join(
+        f"{nm}={engine.stream.kernel_ns(i) / args.steps / 1e6:.3f}ms"
+        f"(x{engine.stream.kernel_launches(i) // max(1, args.steps)})"
+        for i, nm in enumerate(tag_names)))
+
     # roofline for the dominant (forward) kernel, from HIP events on the
     # launching stream; at N=1 there is exactly one fwd launch per step.
     fwd_ns = engine.stream.kernel_ns(shim.KTAG_FWD)

@@ -87,18 +87,23 @@ int main(int argc, char **argv) {
     auto xc = x.cpu().contiguous();
     const float *xp = xc.data_ptr<float>();
     const float *yp = y.data_ptr<float>();
+    /* harness tolerance: |err| <= 1e-4*|ref| + 1e-5 (north_star 1e-4 rel
+     * fp32, absolute floor for cancellation-small outputs) */
+    int bad = 0;
     double worst = 0;
     for (VertexId d = 0; d < V; d++)
       for (int j = 0; j < 8; j++) {
         double acc = 0;
         for (uint32_t e = col_off[d]; e < col_off[d + 1]; e++)
           acc += (double)wf[e] * xp[(int64_t)rows[e] * 8 + j];
-        double err = std::abs(acc - yp[(int64_t)d * 8 + j]) /
-                     (std::abs(acc) + 1e-5);
-        if (err > worst) worst = err;
+        double err = std::abs(acc - yp[(int64_t)d * 8 + j]);
+        if (err > 1e-4 * std::abs(acc) + 1e-5) bad++;
+        double rel = err / (std::abs(acc) + 1e-5);
+        if (rel > worst) worst = rel;
       }
-    printf("aggregation parity vs naive CPU: worst rel err %.3e\n", worst);
-    if (worst > 1e-4) { fprintf(stderr, "PARITY FAIL\n"); return 1; }
+    printf("aggregation parity vs naive CPU: worst rel err %.3e, out-of-tol %d\n",
+           worst, bad);
+    if (bad) { fprintf(stderr, "PARITY FAIL\n"); return 1; }
   }
 
   /* training: features, labels, weights */
