@@ -83,6 +83,8 @@ struct nts_stream {
   double acc_ns[NTS_KTAG_COUNT] = {};
   long long acc_n[NTS_KTAG_COUNT] = {};
   ItemsBuf items_scratch;  /* per-call work-item buffer (stream-ordered) */
+  float *sums_ws = nullptr;  /* per-dst reduction scratch (softmax) */
+  uint64_t sums_cap = 0;
 };
 
 namespace {
@@ -357,22 +359,29 @@ __global__ void k_rows(float *__restrict__ dense, float *__restrict__ packed,
 }
 
 /* ---------------- edge-wise (GAT) kernels ---------------- */
-/* wave-per-destination; lane-strided over the column's edges x f slots */
+/* All edge kernels are driven by the same bounded work items as the
+ * gather (<= NTS_SPLIT edges of one destination per item): a wave-per-
+ * destination scheme would serialize power-law hubs (a 1M-edge hub costs
+ * one wave seconds).  Lanes stride the item's (edges x f) elements. */
 enum EdgeOp { E_SCATTER_SRC, E_GATHER_SRC, E_SCATTER_DST, E_GATHER_DST,
               E_SCATTER_GRAD };
 template <EdgeOp OP>
-__global__ void k_edge_op(float *__restrict__ message,
-                          float *__restrict__ vertex_feat,
-                          const uint32_t *__restrict__ row_indices,
-                          const uint32_t *__restrict__ column_offset,
-                          const uint32_t *__restrict__ mirror_index,
-                          uint32_t batch, uint32_t f) {
+__global__ void k_edge_items(const uint4 *__restrict__ items,
+                             const uint32_t *__restrict__ n_items_p,
+                             float *__restrict__ message,
+                             float *__restrict__ vertex_feat,
+                             const uint32_t *__restrict__ row_indices,
+                             const uint32_t *__restrict__ mirror_index,
+                             uint32_t f) {
+  const uint32_t n_items = *n_items_p;
   const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const uint32_t lane = threadIdx.x & 63;
   const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
-  for (uint32_t d = wave; d < batch; d += n_waves) {
-    const uint32_t e0 = column_offset[d], e1 = column_offset[d + 1];
-    const uint64_t nel = (uint64_t)(e1 - e0) * f;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y;
+    const uint64_t nel = (uint64_t)itm.z * f;
     for (uint64_t i = lane; i < nel; i += 64) {
       const uint32_t e = e0 + (uint32_t)(i / f);
       const uint32_t r = (uint32_t)(i % f);
@@ -389,63 +398,96 @@ __global__ void k_edge_op(float *__restrict__ message,
   }
 }
 
-/* per-edge dot: out[e] = dot(dst_rows[d], src_rows[row_indices[e]-src_s]),
- * one wavefront per destination, lanes strided over the feature dim. */
-__global__ void k_edge_dot(float *__restrict__ out,
-                           const float *__restrict__ dst_rows,
-                           const float *__restrict__ src_rows,
-                           const uint32_t *__restrict__ row_indices,
-                           const uint32_t *__restrict__ column_offset,
-                           uint32_t src_start, uint32_t batch, uint32_t f) {
+/* edge softmax, two item-parallel passes over per-dst partial sums:
+ * pass 1: out[e] = exp(in[e]) (fwd) or g*s (bwd), item-partial sums
+ *         reduced once into sums[dst*f + r] by fp32 atomics;
+ * pass 2: out[e] = exp/sum (fwd) or g*s - sum*s (bwd). */
+template <bool BACKWARD>
+__global__ void k_edge_softmax_sum(const uint4 *__restrict__ items,
+                                   const uint32_t *__restrict__ n_items_p,
+                                   float *__restrict__ out,
+                                   const float *__restrict__ in,
+                                   const float *__restrict__ cached,
+                                   float *__restrict__ sums, uint32_t f) {
+  const uint32_t n_items = *n_items_p;
   const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const uint32_t lane = threadIdx.x & 63;
   const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
-  for (uint32_t d = wave; d < batch; d += n_waves) {
-    const uint32_t e0 = column_offset[d], e1 = column_offset[d + 1];
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y;
+    for (uint32_t r = 0; r < f; ++r) {
+      float part = 0.f;
+      for (uint32_t k = lane; k < itm.z; k += 64) {
+        const uint64_t m = (uint64_t)(e0 + k) * f + r;
+        float v;
+        if (BACKWARD) v = in[m] * cached[m];
+        else v = __expf(in[m]);
+        out[m] = v;          /* stash pass-1 value; normalized in pass 2 */
+        part += v;
+      }
+#pragma unroll
+      for (int w = 32; w >= 1; w >>= 1) part += __shfl_xor(part, w, 64);
+      if (lane == 0) atomicAdd(&sums[(uint64_t)d * f + r], part);
+    }
+  }
+}
+
+template <bool BACKWARD>
+__global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
+                                    const uint32_t *__restrict__ n_items_p,
+                                    float *__restrict__ out,
+                                    const float *__restrict__ cached,
+                                    const float *__restrict__ sums,
+                                    uint32_t f) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y;
+    const uint64_t nel = (uint64_t)itm.z * f;
+    for (uint64_t i = lane; i < nel; i += 64) {
+      const uint32_t e = e0 + (uint32_t)(i / f);
+      const uint32_t r = (uint32_t)(i % f);
+      const uint64_t m = (uint64_t)e * f + r;
+      const float sum = sums[(uint64_t)d * f + r];
+      if (BACKWARD) out[m] = out[m] - sum * cached[m];
+      else out[m] = out[m] / sum;
+    }
+  }
+}
+
+/* per-edge dot: out[e] = dot(dst_rows[d], src_rows[row_indices[e]-src_s]),
+ * item-driven (bounded per-wave work even on power-law hubs); lanes stride
+ * the feature dim per edge. */
+__global__ void k_edge_dot(const uint4 *__restrict__ items,
+                           const uint32_t *__restrict__ n_items_p,
+                           float *__restrict__ out,
+                           const float *__restrict__ dst_rows,
+                           const float *__restrict__ src_rows,
+                           const uint32_t *__restrict__ row_indices,
+                           uint32_t src_start, uint32_t f) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y, cnt = itm.z;
     const float *a = dst_rows + (uint64_t)d * f;
-    for (uint32_t e = e0; e < e1; ++e) {
+    for (uint32_t k = 0; k < cnt; ++k) {
+      const uint32_t e = e0 + k;
       const float *b = src_rows + (uint64_t)(row_indices[e] - src_start) * f;
       float sum = 0.f;
       for (uint32_t j = lane; j < f; j += 64) sum += a[j] * b[j];
 #pragma unroll
       for (int w = 32; w >= 1; w >>= 1) sum += __shfl_xor(sum, w, 64);
       if (lane == 0) out[e] = sum;
-    }
-  }
-}
-
-/* per-dst softmax over incident edge values, per feature slot.
- * Restates edge_softmax_forward_block (ntsCUDADistKernel.cuh:166-213):
- * out[e] = exp(in[e]) / sum_{e' in dst} exp(in[e']), no max subtraction
- * (the exercised GAT config clamps attention scores with leaky_relu before
- * this, keeping exp in range); cached = out for the backward pass. */
-template <bool BACKWARD>
-__global__ void k_edge_softmax(float *__restrict__ out,
-                               const float *__restrict__ in,
-                               const float *__restrict__ cached,
-                               const uint32_t *__restrict__ column_offset,
-                               uint32_t batch, uint32_t f) {
-  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const uint32_t lane = threadIdx.x & 63;
-  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
-  for (uint32_t d = wave; d < batch; d += n_waves) {
-    const uint32_t e0 = column_offset[d], e1 = column_offset[d + 1];
-    for (uint32_t r = 0; r < f; ++r) {
-      float sum = 0.f;
-      for (uint32_t e = e0 + lane; e < e1; e += 64) {
-        const uint64_t m = (uint64_t)e * f + r;
-        if (BACKWARD) sum += in[m] * cached[m];
-        else sum += __expf(in[m]);
-      }
-#pragma unroll
-      for (int w = 32; w >= 1; w >>= 1) sum += __shfl_xor(sum, w, 64);
-      for (uint32_t e = e0 + lane; e < e1; e += 64) {
-        const uint64_t m = (uint64_t)e * f + r;
-        if (BACKWARD)
-          out[m] = in[m] * cached[m] - sum * cached[m];
-        else
-          out[m] = __expf(in[m]) / sum;
-      }
     }
   }
 }
@@ -577,6 +619,7 @@ void nts_stream_destroy(nts_stream *s) {
   }
   if (s->items_scratch.items) hipFree(s->items_scratch.items);
   if (s->items_scratch.counter) hipFree(s->items_scratch.counter);
+  if (s->sums_ws) hipFree(s->sums_ws);
   if (s->owned) hipStreamDestroy(s->stream);
   delete s;
 }
@@ -732,18 +775,41 @@ void nts_scatter_add_rows(nts_stream *s, float *dense, const float *packed,
                      feature_size);
 }
 
+static uint32_t read_edge_count(nts_stream *s, const nts_vid *column_offset,
+                                nts_vid batch) {
+  uint32_t last = 0;
+  NTS_CHECK(hipMemcpyAsync(&last, column_offset + batch, sizeof(uint32_t),
+                           hipMemcpyDeviceToHost, s->stream));
+  NTS_CHECK(hipStreamSynchronize(s->stream));
+  return last;
+}
+
+static float *get_sums(nts_stream *s, uint64_t n) {
+  if (s->sums_cap < n) {
+    NTS_CHECK(hipStreamSynchronize(s->stream));
+    if (s->sums_ws) NTS_CHECK(hipFree(s->sums_ws));
+    NTS_CHECK(hipMalloc(&s->sums_ws, n * sizeof(float)));
+    s->sums_cap = n;
+  }
+  NTS_CHECK(hipMemsetAsync(s->sums_ws, 0, n * sizeof(float), s->stream));
+  return s->sums_ws;
+}
+
 static void launch_edge_op(nts_stream *s, EdgeOp op, float *message,
                            float *vfeat, const nts_vid *row_indices,
                            const nts_vid *column_offset,
                            const nts_vid *mirror_index, nts_vid batch,
                            nts_vid f) {
   if (!batch || !f) return;
+  const uint32_t edges = read_edge_count(s, column_offset, batch);
+  if (!edges) return;
+  ItemsBuf &ib = get_items(s, column_offset, batch, edges);
   Tic t(s, NTS_KTAG_EDGE);
-  const uint32_t grid = grid_for((uint64_t)batch * 64);
+  const uint32_t grid = grid_for(((uint64_t)batch + edges / NTS_SPLIT) * 64);
 #define NTS_ELAUNCH(OP)                                                       \
-  hipLaunchKernelGGL((k_edge_op<OP>), dim3(grid), dim3(NTS_BLOCK), 0,         \
-                     s->stream, message, vfeat, row_indices, column_offset,   \
-                     mirror_index, batch, f)
+  hipLaunchKernelGGL((k_edge_items<OP>), dim3(grid), dim3(NTS_BLOCK), 0,      \
+                     s->stream, ib.items, ib.counter, message, vfeat,         \
+                     row_indices, mirror_index, f)
   switch (op) {
     case E_SCATTER_SRC: NTS_ELAUNCH(E_SCATTER_SRC); break;
     case E_GATHER_SRC: NTS_ELAUNCH(E_GATHER_SRC); break;
@@ -752,6 +818,36 @@ static void launch_edge_op(nts_stream *s, EdgeOp op, float *message,
     case E_SCATTER_GRAD: NTS_ELAUNCH(E_SCATTER_GRAD); break;
   }
 #undef NTS_ELAUNCH
+  dbg_sync(s, "k_edge_items");
+}
+
+static void launch_edge_softmax(nts_stream *s, bool backward, float *out,
+                                const float *in, const float *cached,
+                                const nts_vid *column_offset, nts_vid batch,
+                                nts_vid f) {
+  if (!batch || !f) return;
+  const uint32_t edges = read_edge_count(s, column_offset, batch);
+  if (!edges) return;
+  ItemsBuf &ib = get_items(s, column_offset, batch, edges);
+  float *sums = get_sums(s, (uint64_t)batch * f);
+  const uint32_t grid = grid_for(((uint64_t)batch + edges / NTS_SPLIT) * 64);
+  Tic t(s, NTS_KTAG_EDGE);
+  if (backward) {
+    hipLaunchKernelGGL((k_edge_softmax_sum<true>), dim3(grid), dim3(NTS_BLOCK),
+                       0, s->stream, ib.items, ib.counter, out, in, cached,
+                       sums, f);
+    hipLaunchKernelGGL((k_edge_softmax_norm<true>), dim3(grid),
+                       dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
+                       out, cached, sums, f);
+  } else {
+    hipLaunchKernelGGL((k_edge_softmax_sum<false>), dim3(grid),
+                       dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
+                       out, in, cached, sums, f);
+    hipLaunchKernelGGL((k_edge_softmax_norm<false>), dim3(grid),
+                       dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
+                       out, cached, sums, f);
+  }
+  dbg_sync(s, "k_edge_softmax");
 }
 
 void nts_scatter_src_mirror_to_msg(nts_stream *s, float *message,
@@ -798,11 +894,15 @@ void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
                   const nts_vid *column_offset, nts_vid src_start,
                   nts_vid batch_size, nts_vid feature_size) {
   if (!batch_size || !feature_size) return;
+  const uint32_t edges = read_edge_count(s, column_offset, batch_size);
+  if (!edges) return;
+  ItemsBuf &ib = get_items(s, column_offset, batch_size, edges);
   Tic t(s, NTS_KTAG_EDGE);
-  hipLaunchKernelGGL(k_edge_dot, dim3(grid_for((uint64_t)batch_size * 64)),
-                     dim3(NTS_BLOCK), 0, s->stream, out, dst_rows, src_rows,
-                     row_indices, column_offset, src_start, batch_size,
-                     feature_size);
+  const uint32_t grid =
+      grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
+  hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
+                     ib.items, ib.counter, out, dst_rows, src_rows,
+                     row_indices, src_start, feature_size);
   dbg_sync(s, "k_edge_dot");
 }
 
@@ -823,26 +923,16 @@ void nts_edge_softmax_forward(nts_stream *s, float *msg_output,
                               const nts_vid *column_offset, nts_vid batch_size,
                               nts_vid feature_size) {
   (void)row_indices;
-  if (!batch_size || !feature_size) return;
-  {
-    Tic t(s, NTS_KTAG_EDGE);
-    hipLaunchKernelGGL((k_edge_softmax<false>),
-                       dim3(grid_for((uint64_t)batch_size * 64)),
-                       dim3(NTS_BLOCK), 0, s->stream, msg_output, msg_input,
-                       nullptr, column_offset, batch_size, feature_size);
-  }
+  launch_edge_softmax(s, false, msg_output, msg_input, nullptr, column_offset,
+                      batch_size, feature_size);
   /* cache = output for backward (reference caches at
-   * ntsCUDADistKernel.cuh:210) — one copy kernel; edge count comes from the
-   * offset array's last entry which the host layer knows. */
-  if (msg_cached && msg_cached != msg_output) {
-    uint32_t last = 0;
-    NTS_CHECK(hipMemcpyAsync(&last, column_offset + batch_size,
-                             sizeof(uint32_t), hipMemcpyDeviceToHost,
-                             s->stream));
-    NTS_CHECK(hipStreamSynchronize(s->stream));
-    const uint64_t n_edges_total = (uint64_t)last * feature_size;
-    hipLaunchKernelGGL(k_copy, dim3(grid_for(n_edges_total)), dim3(NTS_BLOCK),
-                       0, s->stream, msg_cached, msg_output, n_edges_total);
+   * ntsCUDADistKernel.cuh:210) */
+  if (msg_cached && msg_cached != msg_output && batch_size && feature_size) {
+    const uint32_t edges = read_edge_count(s, column_offset, batch_size);
+    const uint64_t n = (uint64_t)edges * feature_size;
+    if (n)
+      hipLaunchKernelGGL(k_copy, dim3(grid_for(n)), dim3(NTS_BLOCK), 0,
+                         s->stream, msg_cached, msg_output, n);
   }
 }
 
@@ -853,13 +943,8 @@ void nts_edge_softmax_backward(nts_stream *s, float *msg_input_grad,
                                const nts_vid *column_offset,
                                nts_vid batch_size, nts_vid feature_size) {
   (void)row_indices;
-  if (!batch_size || !feature_size) return;
-  Tic t(s, NTS_KTAG_EDGE);
-  hipLaunchKernelGGL((k_edge_softmax<true>),
-                     dim3(grid_for((uint64_t)batch_size * 64)),
-                     dim3(NTS_BLOCK), 0, s->stream, msg_input_grad,
-                     msg_output_grad, msg_cached, column_offset, batch_size,
-                     feature_size);
+  launch_edge_softmax(s, true, msg_input_grad, msg_output_grad, msg_cached,
+                      column_offset, batch_size, feature_size);
 }
 
 int nts_device_count(void) {
