@@ -44,6 +44,19 @@ def log(msg):
 def build_workload(args):
     from neutronstarlite_amd import graph as G
     t0 = time.time()
+    if args.cfg:
+        from neutronstarlite_amd.config import read_cfg
+        info = read_cfg(args.cfg)
+        path = info.edge_file
+        if not os.path.isabs(path):
+            path = os.path.join(os.path.dirname(os.path.abspath(args.cfg)),
+                                path)
+        edges = G.load_gemini_edges(path)
+        v = info.vertices
+        outd, ind = G.degrees(edges, v)
+        w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+        log(f"cfg workload: {args.cfg} V={v} E={len(edges)}")
+        return v, edges, w
     if args.graph == "reddit":
         v, e = 232_965, 114_000_000
     elif args.graph == "rmat26":
@@ -115,7 +128,13 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--feat", type=int, default=602)
+    ap.add_argument("--feat", type=int, default=None,
+                    help="feature width (default 602, or LAYERS[0] of --cfg)")
+    ap.add_argument("--cfg", default=None,
+                    help="reference-format cfg file (InputInfo KEY:VALUE): "
+                         "takes VERTICES, LAYERS and EDGE_FILE (Gemini "
+                         "binary, resolved relative to the cfg) as the "
+                         "workload instead of --graph")
     ap.add_argument("--model", default="gcn",
                     choices=["gcn", "gat", "gcn-layer"],
                     help="gcn = fused norm-degree aggregation (configs #2-4);"
@@ -140,6 +159,14 @@ def main():
                     help="measured HBM bytes per forward launch from a "
                          "rocprofv3 --pmc pass (see profiles/)")
     args = ap.parse_args()
+
+    if args.feat is None:
+        if args.cfg:
+            from neutronstarlite_amd.config import read_cfg
+            ls = read_cfg(args.cfg).layer_sizes
+            args.feat = ls[0] if ls else 602
+        else:
+            args.feat = 602
 
     if args.traffic_bytes_per_launch is None:
         # measured HBM bytes/launch from the separate rocprofv3 --pmc passes
