@@ -368,7 +368,8 @@ def main():
             "data": "synthetic",
             "config": {
                 "workload": (
-                    f"{args.graph}: RMAT V={v} E={e_total} feat={f} "
+                    (f"cfg {os.path.basename(args.cfg)}: " if args.cfg
+                     else f"{args.graph}: RMAT ") + f"V={v} E={e_total} feat={f} "
                     + ("GAT layer (edge softmax + attention-weighted "
                        "aggregation) fwd+bwd (BASELINE config #5)"
                        if args.model == "gat" else
