@@ -173,7 +173,7 @@ def main():
         # committed under profiles/ (FETCH corrected x2 per the gfx950
         # wide-read calibration + WRITE), keyed by exact profiled config
         args.traffic_bytes_per_launch = {
-            ("reddit", 602, "none", "gcn"): 2.62e11,  # profiles/round1
+            ("reddit", 602, "none", "gcn"): 2.64e11,  # profiles/round1 (final PMC)
         }.get((args.graph, args.feat, args.relabel, args.model))
 
     import torch

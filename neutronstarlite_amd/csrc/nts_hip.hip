@@ -60,7 +60,7 @@ uint32_t nts_split() {
   return v;
 }
 uint32_t nts_max_blocks() {
-  static uint32_t v = env_u32("NTS_MAX_BLOCKS", 4096);
+  static uint32_t v = env_u32("NTS_MAX_BLOCKS", 8192);
   return v;
 }
 #define NTS_SPLIT nts_split()
