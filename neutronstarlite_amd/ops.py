@@ -89,6 +89,48 @@ class HipEngine:
             ch.edge_size, ch.src_n, f, with_weight)
 
 
+class MiniBatchFuseOp:
+    """MiniBatchFuseOp equivalent (core/ntsMiniBatchGraphOp.hpp:61-131):
+    the same aggregation arithmetic on one sampled layer's compacted
+    subgraph — forward pulls compacted source rows into destination rows,
+    backward pushes destination grads back to compacted sources.  Runs the
+    same gfx950 gather kernels with local ids (src range [0, n_src))."""
+
+    def __init__(self, layer, device, engine: "HipEngine" = None):
+        self.layer = layer
+        self.engine = engine or HipEngine()
+        self.column_offset = _u32_cuda(layer.column_offset, device)
+        self.row_indices = _u32_cuda(layer.row_indices_local, device)
+        self.w_fwd = torch.from_numpy(layer.edge_weight).to(device)
+        self.row_offset = _u32_cuda(layer.row_offset, device)
+        self.column_indices = _u32_cuda(layer.column_indices_local, device)
+        self.w_bwd = torch.from_numpy(layer.edge_weight_backward).to(device)
+
+    def forward(self, x_compact: torch.Tensor) -> torch.Tensor:
+        ly = self.layer
+        assert x_compact.shape[0] == ly.n_src and x_compact.is_contiguous()
+        f = x_compact.shape[1]
+        y = torch.zeros(ly.n_dst, f, dtype=torch.float32,
+                        device=x_compact.device)
+        self.engine.stream.gather_by_dst_from_src(
+            x_compact.data_ptr(), y.data_ptr(), self.w_fwd.data_ptr(),
+            self.row_indices.data_ptr(), self.column_offset.data_ptr(),
+            0, ly.n_src, 0, ly.n_dst, ly.e_size, ly.n_dst, f, True)
+        return y
+
+    def backward(self, grad_y: torch.Tensor) -> torch.Tensor:
+        ly = self.layer
+        assert grad_y.shape[0] == ly.n_dst and grad_y.is_contiguous()
+        f = grad_y.shape[1]
+        gx = torch.zeros(ly.n_src, f, dtype=torch.float32,
+                         device=grad_y.device)
+        self.engine.stream.gather_by_src_from_dst(
+            grad_y.data_ptr(), gx.data_ptr(), self.w_bwd.data_ptr(),
+            self.row_offset.data_ptr(), self.column_indices.data_ptr(),
+            0, ly.n_src, 0, ly.n_dst, ly.e_size, ly.n_src, f, True)
+        return gx
+
+
 class SingleGPUFuseOp:
     """ForwardSingleGPUfuseOp equivalent: whole graph as one chunk on one GPU
     (core/ntsSingleGPUFusedGraphOp.hpp:48-71)."""
