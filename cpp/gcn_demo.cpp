@@ -137,6 +137,36 @@ int main(int argc, char **argv) {
   }
   printf("loss %0.4f -> %0.4f over %d epochs\n", first_loss, last_loss, epochs);
   if (!(last_loss < first_loss)) { fprintf(stderr, "NO LEARNING\n"); return 1; }
+
+  /* GAT 5-op chain through the decomposed surface (the call shape of
+   * toolkits/GAT_GPU_DIST.hpp:191-215): per-edge attention scalars ->
+   * softmax -> aggregate; softmax rows must sum to 1 per destination. */
+  {
+    NtsVar s_src = torch::rand({(int64_t)V, 1}, torch::device(dev)) * 2 - 1;
+    NtsVar s_dst = torch::rand({(int64_t)V, 1}, torch::device(dev)) * 2 - 1;
+    op::DistGPUGetDepNbrOp dep(&pg, &active);
+    op::DistGPUScatterSrc ssrc(&pg, &active);
+    op::DistGPUScatterDst sdst(&pg, &active);
+    op::DistGPUEdgeSoftMax smax(&pg, &active);
+    op::DistGPUAggregateDst aggd(&pg, &active);
+    NtsVar mirror = dep.forward(s_src);
+    NtsVar m1 = ssrc.forward(mirror);
+    NtsVar m2 = sdst.forward(s_dst);
+    NtsVar e_att = torch::leaky_relu(m1 + m2, 0.2);
+    NtsVar a = smax.forward(e_att);
+    NtsVar ones_per_dst = aggd.forward(a);   /* sum of softmax per dst */
+    auto sums = ones_per_dst.cpu();
+    const float *sp = sums.data_ptr<float>();
+    int badsm = 0;
+    for (VertexId d = 0; d < V; d++) {
+      const bool has_edges = col_off[d + 1] > col_off[d];
+      if (has_edges && std::abs(sp[d] - 1.0f) > 1e-4f) badsm++;
+    }
+    NtsVar ag = smax.backward(a);  /* exercise backward path */
+    (void)ag;
+    printf("gat 5-op chain: softmax rows off-1: %d\n", badsm);
+    if (badsm) { fprintf(stderr, "GAT CHAIN FAIL\n"); return 1; }
+  }
   printf("gcn_demo ok\n");
   return 0;
 }

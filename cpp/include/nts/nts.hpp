@@ -168,6 +168,173 @@ class ForwardSingleGPUfuseOp : public ntsGraphOp {
   }
 };
 
+/* ---- decomposed edge-valued ops (GAT path, config #5) ----
+ * The class names and signatures of core/ntsDistGPUGraphOp.hpp:48-361, so
+ * GAT_GPU_DIST.hpp's 5-op chain (:191-215) compiles unchanged.  This host
+ * layer implements the single-partition case (every vertex its own master;
+ * mirror_index = identity — PartitionedGraph::generateMirrorIndex,
+ * PartitionedGraph.hpp:295-305); the multi-partition mirror exchange lives
+ * in the RCCL ring driver. */
+
+namespace detail {
+inline torch::Tensor &identity_mirror_index(PartitionedGraph *pg,
+                                            torch::Device dev) {
+  static torch::Tensor idx;  /* single-GPU demo scope: whole-graph identity */
+  const int64_t v = pg->partition_offset.back();
+  if (!idx.defined() || idx.size(0) != v || idx.device() != dev) {
+    idx = torch::arange(v, torch::TensorOptions()
+                               .dtype(torch::kInt32).device(dev));
+  }
+  return idx;
+}
+}  // namespace detail
+
+/* gather dependent-neighbor features into the mirror matrix: with one
+ * partition this is the identity over the input (DistGPUGetDepNbrOp,
+ * ntsDistGPUGraphOp.hpp:48-143 — comm path degenerates). */
+class DistGPUGetDepNbrOp : public ntsGraphOp {
+ public:
+  DistGPUGetDepNbrOp(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {}
+  NtsVar forward(NtsVar &f_input) override { return f_input.clone(); }
+  NtsVar backward(NtsVar &output_grad) override { return output_grad.clone(); }
+};
+
+/* per-edge materialize of source-mirror rows (DistGPUScatterSrc,
+ * ntsDistGPUGraphOp.hpp:145-196). */
+class DistGPUScatterSrc : public ntsGraphOp {
+ public:
+  DistGPUScatterSrc(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {}
+  NtsVar forward(NtsVar &mirror) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    auto &mi = detail::identity_mirror_index(partitioned_graph_,
+                                             mirror.device());
+    NtsVar msg = torch::zeros({(int64_t)c.edge_size, mirror.size(1)},
+                              mirror.options());
+    nts_scatter_src_mirror_to_msg(
+        partitioned_graph_->stream, msg.data_ptr<float>(),
+        mirror.contiguous().data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(),
+        (const uint32_t *)mi.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)mirror.size(1));
+    return msg;
+  }
+  NtsVar backward(NtsVar &msg_grad) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    auto &mi = detail::identity_mirror_index(partitioned_graph_,
+                                             msg_grad.device());
+    NtsVar g = torch::zeros({(int64_t)c.src_n(), msg_grad.size(1)},
+                            msg_grad.options());
+    nts_gather_msg_to_src_mirror(
+        partitioned_graph_->stream, g.data_ptr<float>(),
+        msg_grad.contiguous().data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(),
+        (const uint32_t *)mi.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)msg_grad.size(1));
+    return g;
+  }
+};
+
+/* per-edge materialize of destination rows (DistGPUScatterDst,
+ * ntsDistGPUGraphOp.hpp:198-248). */
+class DistGPUScatterDst : public ntsGraphOp {
+ public:
+  DistGPUScatterDst(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {}
+  NtsVar forward(NtsVar &dst_feat) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar msg = torch::zeros({(int64_t)c.edge_size, dst_feat.size(1)},
+                              dst_feat.options());
+    nts_scatter_dst_to_msg(
+        partitioned_graph_->stream, msg.data_ptr<float>(),
+        dst_feat.contiguous().data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)dst_feat.size(1));
+    return msg;
+  }
+  NtsVar backward(NtsVar &msg_grad) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar g = torch::zeros({(int64_t)c.dst_n(), msg_grad.size(1)},
+                            msg_grad.options());
+    nts_gather_msg_to_dst(
+        partitioned_graph_->stream, g.data_ptr<float>(),
+        msg_grad.contiguous().data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)msg_grad.size(1));
+    return g;
+  }
+};
+
+/* edge->destination reduce (DistGPUAggregateDst,
+ * ntsDistGPUGraphOp.hpp:250-300). */
+class DistGPUAggregateDst : public ntsGraphOp {
+ public:
+  DistGPUAggregateDst(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {}
+  NtsVar forward(NtsVar &msg) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar y = torch::zeros({(int64_t)c.dst_n(), msg.size(1)},
+                            msg.options());
+    nts_gather_msg_to_dst(
+        partitioned_graph_->stream, y.data_ptr<float>(),
+        msg.contiguous().data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)msg.size(1));
+    return y;
+  }
+  NtsVar backward(NtsVar &y_grad) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar mg = torch::zeros({(int64_t)c.edge_size, y_grad.size(1)},
+                             y_grad.options());
+    nts_scatter_grad_back_to_message(
+        partitioned_graph_->stream, y_grad.contiguous().data_ptr<float>(),
+        mg.data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)y_grad.size(1));
+    return mg;
+  }
+};
+
+/* per-destination softmax over incident edge values (DistGPUEdgeSoftMax,
+ * ntsDistGPUGraphOp.hpp:302-361; no max subtraction, cached for backward). */
+class DistGPUEdgeSoftMax : public ntsGraphOp {
+  NtsVar cached_;
+ public:
+  DistGPUEdgeSoftMax(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {}
+  NtsVar forward(NtsVar &msg) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar out = torch::zeros_like(msg);
+    cached_ = torch::zeros_like(msg);
+    nts_edge_softmax_forward(
+        partitioned_graph_->stream, out.data_ptr<float>(),
+        msg.contiguous().data_ptr<float>(), cached_.data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)msg.size(1));
+    return out;
+  }
+  NtsVar backward(NtsVar &out_grad) override {
+    auto &c = *partitioned_graph_->graph_chunks[0];
+    NtsVar g = torch::zeros_like(out_grad);
+    nts_edge_softmax_backward(
+        partitioned_graph_->stream, g.data_ptr<float>(),
+        out_grad.contiguous().data_ptr<float>(),
+        cached_.data_ptr<float>(),
+        (const uint32_t *)c.row_indices.data_ptr<int32_t>(),
+        (const uint32_t *)c.column_offset.data_ptr<int32_t>(), c.dst_n(),
+        (uint32_t)out_grad.size(1));
+    return g;
+  }
+};
+
 }  // namespace op
 
 /* Tape-based context (NtsContext surface, core/ntsContext.hpp:108-359):
