@@ -480,7 +480,29 @@ __global__ void k_edge_dot(const uint4 *__restrict__ items,
     const uint32_t d = itm.x & 0x7fffffffu;
     const uint32_t e0 = itm.y, cnt = itm.z;
     const float *a = dst_rows + (uint64_t)d * f;
-    for (uint32_t k = 0; k < cnt; ++k) {
+    /* 4-edge unroll: four independent source-row loads in flight */
+    uint32_t k = 0;
+    for (; k + 4 <= cnt; k += 4) {
+      const float *b[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        b[q] = src_rows +
+               (uint64_t)(row_indices[e0 + k + q] - src_start) * f;
+      float sum[4] = {};
+      for (uint32_t j = lane; j < f; j += 64) {
+        const float av = a[j];
+#pragma unroll
+        for (int q = 0; q < 4; ++q) sum[q] = fmaf(av, b[q][j], sum[q]);
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+#pragma unroll
+        for (int w = 32; w >= 1; w >>= 1)
+          sum[q] += __shfl_xor(sum[q], w, 64);
+        if (lane == 0) out[e0 + k + q] = sum[q];
+      }
+    }
+    for (; k < cnt; ++k) {
       const uint32_t e = e0 + k;
       const float *b = src_rows + (uint64_t)(row_indices[e] - src_start) * f;
       float sum = 0.f;

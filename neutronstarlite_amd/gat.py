@@ -47,6 +47,7 @@ class GATLayer:
         perm = np.argsort(self._src_of_edge_np, kind="stable")
         self.csr_from_csc = torch.from_numpy(perm).to(device)
         self._doe = self._soe = None
+        self._ones_dst = self._ones_src = None
 
     @property
     def dst_of_edge(self):
@@ -122,14 +123,26 @@ class GATLayer:
                                  ch.column_offset.data_ptr(), ch.dst_n, 1)
         ge = ge * torch.where(saved["m_sum"] > 0, 1.0, negative_slope)
         ge = ge.contiguous()
+        # edge-scalar -> vertex reductions through the load-balanced gather
+        # kernel (per-edge values as WEIGHTS over an all-ones input): the
+        # direct msg->vertex atomics serialize on power-law hub sources
+        # (one address takes every hub edge's atomicAdd — measured 25 ms of
+        # a 115 ms step), while the gather's work items split hubs and merge
+        # a handful of partials.
+        if self._ones_dst is None:
+            self._ones_dst = torch.ones(ch.dst_n, 1, device=dev)
+            self._ones_src = torch.ones(ch.src_n, 1, device=dev)
+        ge_csr = ge[self.csr_from_csc].contiguous()
         g_src = torch.zeros(ch.src_n, 1, device=dev)
-        st.gather_msg_to_src_mirror(g_src.data_ptr(), ge.data_ptr(),
-                                    ch.row_indices.data_ptr(),
-                                    ch.column_offset.data_ptr(),
-                                    self.mirror_index.data_ptr(),
-                                    ch.dst_n, 1)
+        st.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),
+                                  ge_csr.data_ptr(), ch.row_offset.data_ptr(),
+                                  ch.column_indices.data_ptr(),
+                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
+                                  E, ch.src_n, 1, with_weight=True)
         g_dst = torch.zeros(ch.dst_n, 1, device=dev)
-        st.gather_msg_to_dst(g_dst.data_ptr(), ge.data_ptr(),
-                             ch.row_indices.data_ptr(),
-                             ch.column_offset.data_ptr(), ch.dst_n, 1)
+        st.gather_by_dst_from_src(self._ones_src.data_ptr(), g_dst.data_ptr(),
+                                  ge.data_ptr(), ch.row_indices.data_ptr(),
+                                  ch.column_offset.data_ptr(),
+                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
+                                  E, ch.dst_n, 1, with_weight=True)
         return grad_h, g_src[:, 0], g_dst[:, 0]
