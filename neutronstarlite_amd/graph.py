@@ -19,7 +19,7 @@ neighbor-aggregation path (all citations into /root/reference):
 This module is pure host-side plumbing (numpy); the compute path is the HIP
 extension (csrc/nts_hip.hip behind include/nts_hip.h).
 """
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List
 
 import numpy as np

@@ -42,7 +42,6 @@ def test_shim_binds_and_reports_arch():
 
 
 def test_shim_fails_loudly_when_missing(monkeypatch):
-    import importlib
     import neutronstarlite_amd.shim as shim
     monkeypatch.setattr(shim, "_SO", "/nonexistent/libnts_hip.so")
     monkeypatch.setattr(shim, "_lib", None)

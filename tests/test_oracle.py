@@ -5,7 +5,6 @@ golden vectors for this path, so these ARE the pin."""
 import os
 
 import numpy as np
-import pytest
 import scipy.sparse as sp
 
 import oracle
