@@ -151,6 +151,10 @@ def main():
     ap.add_argument("--cache-dir", default=None,
                     help="cache built graph/chunk arrays (npz) to skip the "
                          "~1 min host-side setup on repeated runs")
+    ap.add_argument("--mirror-filtered", action="store_true",
+                    help="lock-free exchange: send each peer only the rows "
+                         "its chunks reference (the reference's LOCK_FREE "
+                         "path; static index lists exchanged at setup)")
     ap.add_argument("--relabel", default="none", choices=["none", "degree"],
                     help="preprocessing: renumber vertices by descending "
                          "out-degree so hot source rows are contiguous "
@@ -242,6 +246,9 @@ def main():
     engine = HipEngine()
     engine.stream.timing(True)
     rg = RingGraph(offs, rank, dchunks, dev)
+    if distributed and args.mirror_filtered:
+        from neutronstarlite_amd.ring import setup_mirror_lists
+        setup_mirror_lists(rg)
 
     rng = np.random.default_rng(42 + rank)
     x = torch.from_numpy(

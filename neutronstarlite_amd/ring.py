@@ -55,6 +55,98 @@ def _exchange(send_t, dst, recv_t, src):
     return dist.batch_isend_irecv(ops)
 
 
+def setup_mirror_lists(rg: "RingGraph"):
+    """The reference's lock-free path (LOCK_FREE:1; mirror discovery
+    PartitionedGraph::DetermineMirror, PartitionedGraph.hpp:174-209;
+    filtered emit via forward_multisocket_message_index,
+    ntsCPUFusedGraphOp.hpp:57-69): instead of broadcasting the whole owned
+    block each ring step, each rank sends a peer ONLY the rows that peer's
+    chunk actually references.  The static index lists travel once at setup:
+
+      rg.need[k]  = unique global ids of partition k that OUR chunk k reads
+                    (recv side: which rows we need from rank k)
+      rg.serve[j] = the ids rank j declared needing from US (send side)
+
+    Payloads stay dense fp32 row blocks (packed by index_select /
+    nts_gather_rows); no [vid|floats] records.  Call once after RingGraph
+    construction; ring_forward then runs mirror-filtered automatically."""
+    P, r = rg.world, rg.rank
+    dev = rg.device
+    need = []
+    for k in range(P):
+        ch = rg.chunks[k]
+        if k == r:
+            need.append(None)
+            continue
+        ri = ch.row_indices
+        if isinstance(ri, np.ndarray):       # CPU (gloo-test) chunks
+            t = torch.from_numpy(ri.astype(np.int64))
+        else:                                 # DeviceChunk: u32 as int32
+            t = ri.to(torch.int64) & 0xFFFFFFFF
+        need.append(torch.unique(t))
+    # exchange list lengths, then the lists themselves (setup-time, tiny)
+    lens = torch.zeros(P, P, dtype=torch.int64)
+    for k in range(P):
+        if k != r:
+            lens[r, k] = len(need[k])
+    lens_w = lens.to(dev) if dist.get_backend() == "nccl" else lens
+    dist.all_reduce(lens_w, op=dist.ReduceOp.SUM)
+    lens = lens_w.cpu()
+    serve = [None] * P
+    reqs = []
+    for k in range(P):
+        if k == r:
+            continue
+        # we tell rank k what we need from it; rank k tells us what it needs
+        send_ids = need[k].to(dev)
+        recv_ids = torch.zeros(int(lens[k, r].item()), dtype=torch.int64,
+                               device=dev)
+        reqs += _exchange(send_ids, k, recv_ids, k)
+        serve[k] = recv_ids
+    for rq in reqs:
+        rq.wait()
+    rg.need = need
+    rg.serve = [s if s is None else s.to(dev) for s in serve]
+    rg.mirror_filtered = True
+
+
+def _ring_forward_filtered(rg: "RingGraph", x_owned, engine):
+    """Mirror-filtered forward: per step, pack only the rows the peer
+    declared needing; the receiver scatters them into a dense block over the
+    sender's partition range and aggregates that chunk."""
+    P, r = rg.world, rg.rank
+    f = x_owned.shape[1]
+    dev = x_owned.device
+    y = torch.zeros(rg.owned_n, f, dtype=torch.float32, device=dev)
+    lo = rg.offs[r]
+
+    def pack_for(peer):
+        rows = rg.serve[peer] - lo
+        return x_owned.index_select(0, rows).contiguous()
+
+    def post(step):
+        to = (r - step) % P
+        frm = (r + step) % P
+        recv = torch.empty(len(rg.need[frm]), f, device=dev)
+        reqs = _exchange(pack_for(to), to, recv, frm)
+        return reqs, recv, frm
+
+    pending = post(1)
+    engine.csc_forward(rg.chunks[r], x_owned, y)
+    for step in range(1, P):
+        reqs, recv, frm = pending
+        if step + 1 < P:
+            nxt = post(step + 1)
+        for rq in reqs:
+            rq.wait()
+        dense = torch.zeros(rg.part_n(frm), f, device=dev)
+        dense.index_copy_(0, rg.need[frm] - rg.offs[frm], recv)
+        engine.csc_forward(rg.chunks[frm], dense, y)
+        if step + 1 < P:
+            pending = nxt
+    return y
+
+
 def ring_forward(rg: RingGraph, x_owned: torch.Tensor, engine) -> torch.Tensor:
     """Distributed forward aggregation: returns y over the owned dst range.
     Mirrors sync_compute_decoupled's ring loop (graph.hpp:3685-3707)."""
@@ -65,6 +157,8 @@ def ring_forward(rg: RingGraph, x_owned: torch.Tensor, engine) -> torch.Tensor:
     if P == 1:
         engine.csc_forward(rg.chunks[r], x_owned, y)
         return y
+    if getattr(rg, "mirror_filtered", False):
+        return _ring_forward_filtered(rg, x_owned, engine)
     maxn = max(rg.part_n(k) for k in range(P))
     bufs = [torch.empty(maxn, f, dtype=torch.float32, device=x_owned.device)
             for _ in range(2)]

@@ -44,7 +44,7 @@ def _data(world):
     return edges, w, offs, x, g
 
 
-def _worker(rank, world, tmpdir, q):
+def _worker(rank, world, filtered, tmpdir, q):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         dist.init_process_group(
@@ -53,6 +53,9 @@ def _worker(rank, world, tmpdir, q):
         edges, w, offs, x, g = _data(world)
         chunks = G.build_chunks(edges, w, offs, rank)
         rg = RingGraph(offs, rank, chunks, torch.device("cpu"))
+        if filtered:
+            from neutronstarlite_amd.ring import setup_mirror_lists
+            setup_mirror_lists(rg)
         lo, hi = int(offs[rank]), int(offs[rank + 1])
         eng = OracleEngine()
         y = ring_forward(rg, torch.from_numpy(x[lo:hi]).clone(), eng)
@@ -65,11 +68,12 @@ def _worker(rank, world, tmpdir, q):
 
 
 @pytest.mark.timeout(300)
-@pytest.mark.parametrize("world", [2, 3])
-def test_ring_matches_whole_graph(tmp_path, world):
+@pytest.mark.parametrize("world,filtered", [(2, False), (3, False), (3, True)])
+def test_ring_matches_whole_graph(tmp_path, world, filtered):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_worker, args=(r, world, str(tmp_path), q))
+    procs = [ctx.Process(target=_worker,
+                         args=(r, world, filtered, str(tmp_path), q))
              for r in range(world)]
     for p in procs:
         p.start()
