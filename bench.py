@@ -177,7 +177,10 @@ def main():
         # committed under profiles/ (FETCH corrected x2 per the gfx950
         # wide-read calibration + WRITE), keyed by exact profiled config
         args.traffic_bytes_per_launch = {
-            ("reddit", 602, "none", "gcn"): 2.64e11,  # profiles/round1 (final PMC)
+            # measured per gather launch (FETCH x2 gfx950 correction + WRITE)
+            ("reddit", 602, "none", "gcn"): 2.64e11,  # profiles/round1 final
+            ("reddit", 128, "none", "gcn"): 3.11e10,  # 0.52x model: L3 reuse
+            ("reddit", 256, "none", "gcn"): 7.20e10,  # 0.61x model: L3 reuse
         }.get((args.graph, args.feat, args.relabel, args.model))
 
     import torch
