@@ -123,7 +123,13 @@ def cpu_baseline_leg(chunks, f, target_sec=15.0):
     }
 
 
-def main():
+def main(argv=None, _test_engine_factory=None, _test_backend=None,
+         _test_device=None):
+    """Product entry point: HIP engine, RCCL, cuda devices.  The _test_*
+    hooks exist ONLY so tests/test_bench_dist_cpu.py can drive this exact
+    orchestration (env parsing, process group, barriers, max-over-ranks
+    timing, JSON contract) under gloo on CPU with a test-owned engine; the
+    product defaults never touch them."""
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
@@ -162,7 +168,7 @@ def main():
     ap.add_argument("--traffic-bytes-per-launch", type=float, default=None,
                     help="measured HBM bytes per forward launch from a "
                          "rocprofv3 --pmc pass (see profiles/)")
-    args = ap.parse_args()
+    args = ap.parse_args(argv)
 
     if args.feat is None:
         if args.cfg:
@@ -197,11 +203,18 @@ def main():
     if args.gpus > 1 and world == 1:
         raise SystemExit("--gpus N>1 must be launched via torch.distributed.run")
 
-    assert torch.cuda.is_available(), "bench needs an MI355X"
-    torch.cuda.set_device(local_rank)
-    dev = torch.device("cuda", local_rank)
+    if _test_device == "cpu":
+        dev = torch.device("cpu")
+    else:
+        assert torch.cuda.is_available(), "bench needs an MI355X"
+        torch.cuda.set_device(local_rank)
+        dev = torch.device("cuda", local_rank)
     if distributed:
-        dist.init_process_group("nccl")
+        dist.init_process_group(_test_backend or "nccl")
+
+    def device_sync():
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
 
     f = args.feat
     cache = None
@@ -245,8 +258,12 @@ def main():
 
     e_total = edges_len
     lo, hi = int(offs[rank]), int(offs[rank + 1])
-    dchunks = [DeviceChunk(ch, dev) for ch in chunks]
-    engine = HipEngine()
+    if _test_engine_factory is None:
+        dchunks = [DeviceChunk(ch, dev) for ch in chunks]
+        engine = HipEngine()
+    else:
+        dchunks = chunks
+        engine = _test_engine_factory()
     engine.stream.timing(True)
     rg = RingGraph(offs, rank, dchunks, dev)
     if distributed and args.mirror_filtered:
@@ -309,7 +326,7 @@ def main():
     log(f"rank {rank}/{n}: warmup {args.warmup}")
     for _ in range(args.warmup):
         step()
-    torch.cuda.synchronize()
+    device_sync()
     if distributed:
         dist.barrier()
     engine.stream.timing_reset()
@@ -317,7 +334,7 @@ def main():
     t0 = time.time()
     for _ in range(args.steps):
         step()
-    torch.cuda.synchronize()
+    device_sync()
     if distributed:
         dist.barrier()
     elapsed = time.time() - t0
