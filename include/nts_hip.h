@@ -189,6 +189,22 @@ void nts_scatter_grad_back_to_message(nts_stream *s, const float *input_grad,
     float *message_grad, const nts_vid *row_indices,
     const nts_vid *column_offset, nts_vid batch_size, nts_vid feature_size);
 
+/* ---- GPU-resident neighbor sampling (additive) ----
+ * Reservoir fan-out selection on device over the whole-graph CSC
+ * (reference semantics: Sampler::reservoir_sample, ntsSampler.hpp:113-166 —
+ * keep the first `fanout` edge slots of a column, then slot j >= fanout
+ * replaces a uniformly chosen earlier slot with probability fanout/(j+1)).
+ * For each of the `n_dst` destinations: writes min(deg, fanout) sampled
+ * GLOBAL source ids into out_src[d*fanout ..] and the count into
+ * out_cnt[d].  Deterministic in (seed, column content): the per-step RNG is
+ * a counter hash of (seed, dst, step).  Compaction into a sampCSC-style
+ * local subgraph is host-layer plumbing (torch unique/searchsorted on
+ * device — see neutronstarlite_amd/sampler_gpu.py). */
+void nts_sample_reservoir(nts_stream *s, const nts_vid *column_offset,
+    const nts_vid *row_indices, const nts_vid *dst_list, nts_vid n_dst,
+    nts_vid fanout, unsigned long long seed, nts_vid *out_src,
+    nts_vid *out_cnt);
+
 /* Device info for the host layer / bench. */
 int nts_device_count(void);
 void nts_set_device(int dev);

@@ -310,6 +310,61 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
   }
 }
 
+/* ---- GPU-resident reservoir sampling ----
+ * One wave per destination; reservoir semantics of
+ * Sampler::reservoir_sample (ntsSampler.hpp:113-166): keep the first
+ * `fanout` edge slots, then slot j replaces a uniform earlier slot with
+ * probability fanout/(j+1).  Counter-based splitmix hash keeps the draw
+ * deterministic in (seed, dst, step).  Columns with deg <= fanout copy in
+ * parallel; the (rare) long-column reservoir walk runs on lane 0 with the
+ * reservoir staged in LDS. */
+__device__ __forceinline__ uint32_t k_hash_u32(unsigned long long seed,
+                                               uint32_t d, uint32_t j) {
+  unsigned long long z = seed ^ ((unsigned long long)d << 32) ^ j;
+  z = (z ^ (z >> 33)) * 0xff51afd7ed558ccdULL;
+  z = (z ^ (z >> 33)) * 0xc4ceb9fe1a85ec53ULL;
+  return (uint32_t)(z ^ (z >> 33));
+}
+
+constexpr uint32_t NTS_MAX_FANOUT = 1024;
+
+__global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
+                                   const uint32_t *__restrict__ row_indices,
+                                   const uint32_t *__restrict__ dst_list,
+                                   uint32_t n_dst, uint32_t fanout,
+                                   unsigned long long seed,
+                                   uint32_t *__restrict__ out_src,
+                                   uint32_t *__restrict__ out_cnt) {
+  __shared__ uint32_t s_res[4][NTS_MAX_FANOUT];  /* 4 waves per block */
+  const uint32_t wave_in_block = threadIdx.x >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  uint32_t *res = s_res[wave_in_block];
+  for (uint32_t i = wave; i < n_dst; i += n_waves) {
+    const uint32_t d = dst_list[i];
+    const uint32_t e0 = column_offset[d];
+    const uint32_t deg = column_offset[d + 1] - e0;
+    const uint32_t k = deg < fanout ? deg : fanout;
+    if (deg <= fanout) {
+      for (uint32_t j = lane; j < k; j += 64)
+        out_src[(uint64_t)i * fanout + j] = row_indices[e0 + j];
+    } else {
+      if (lane == 0) {
+        for (uint32_t j = 0; j < fanout; ++j) res[j] = row_indices[e0 + j];
+        for (uint32_t j = fanout; j < deg; ++j) {
+          const uint32_t r = k_hash_u32(seed, d, j) % (j + 1);
+          if (r < fanout) res[r] = row_indices[e0 + j];
+        }
+      }
+      __builtin_amdgcn_wave_barrier();
+      for (uint32_t j = lane; j < k; j += 64)
+        out_src[(uint64_t)i * fanout + j] = res[j];
+    }
+    if (lane == 0) out_cnt[i] = k;
+  }
+}
+
 /* message unpack: records [u32 vid | f x f32], stride f+1 floats */
 __global__ void k_deserialize(const float *__restrict__ msg, uint32_t count,
                               uint32_t part_start, float *__restrict__ dense,
@@ -967,6 +1022,25 @@ void nts_edge_softmax_backward(nts_stream *s, float *msg_input_grad,
   (void)row_indices;
   launch_edge_softmax(s, true, msg_input_grad, msg_output_grad, msg_cached,
                       column_offset, batch_size, feature_size);
+}
+
+void nts_sample_reservoir(nts_stream *s, const nts_vid *column_offset,
+                          const nts_vid *row_indices, const nts_vid *dst_list,
+                          nts_vid n_dst, nts_vid fanout,
+                          unsigned long long seed, nts_vid *out_src,
+                          nts_vid *out_cnt) {
+  if (!n_dst || !fanout) return;
+  if (fanout > NTS_MAX_FANOUT) {
+    fprintf(stderr, "nts_sample_reservoir: fanout %u > %u\n", fanout,
+            NTS_MAX_FANOUT);
+    abort();
+  }
+  Tic t(s, NTS_KTAG_ITEMS);
+  hipLaunchKernelGGL(k_sample_reservoir,
+                     dim3(grid_for((uint64_t)n_dst * 64)), dim3(NTS_BLOCK), 0,
+                     s->stream, column_offset, row_indices, dst_list, n_dst,
+                     fanout, seed, out_src, out_cnt);
+  dbg_sync(s, "k_sample_reservoir");
 }
 
 int nts_device_count(void) {

@@ -99,12 +99,20 @@ class MiniBatchFuseOp:
     def __init__(self, layer, device, engine: "HipEngine" = None):
         self.layer = layer
         self.engine = engine or HipEngine()
-        self.column_offset = _u32_cuda(layer.column_offset, device)
-        self.row_indices = _u32_cuda(layer.row_indices_local, device)
-        self.w_fwd = torch.from_numpy(layer.edge_weight).to(device)
-        self.row_offset = _u32_cuda(layer.row_offset, device)
-        self.column_indices = _u32_cuda(layer.column_indices_local, device)
-        self.w_bwd = torch.from_numpy(layer.edge_weight_backward).to(device)
+
+        def up(a):
+            if isinstance(a, torch.Tensor):       # GPU-resident sampler
+                return a.to(device)
+            if a.dtype == np.uint32:
+                return _u32_cuda(a, device)
+            return torch.from_numpy(a).to(device)
+
+        self.column_offset = up(layer.column_offset)
+        self.row_indices = up(layer.row_indices_local)
+        self.w_fwd = up(layer.edge_weight)
+        self.row_offset = up(layer.row_offset)
+        self.column_indices = up(layer.column_indices_local)
+        self.w_bwd = up(layer.edge_weight_backward)
 
     def forward(self, x_compact: torch.Tensor) -> torch.Tensor:
         ly = self.layer

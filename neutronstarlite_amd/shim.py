@@ -87,6 +87,8 @@ def lib():
     l.nts_edge_dot.argtypes = [_vp] + [_vp] * 5 + [_u32] * 3
     l.nts_edge_softmax_forward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_sample_reservoir.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32,
+                                       _c.c_ulonglong, _vp, _vp]
     l.nts_device_count.restype = _i32
     l.nts_set_device.argtypes = [_i32]
     l.nts_build_arch.restype = _c.c_char_p
@@ -236,6 +238,13 @@ class Stream:
         self._lib.nts_edge_softmax_backward(
             self.h, _vp(msg_in_grad), _vp(msg_out_grad), _vp(msg_cached),
             _vp(row_indices), _vp(column_offset), batch, f)
+
+    def sample_reservoir(self, column_offset, row_indices, dst_list, n_dst,
+                         fanout, seed, out_src, out_cnt):
+        self._lib.nts_sample_reservoir(self.h, _vp(column_offset),
+                                       _vp(row_indices), _vp(dst_list),
+                                       n_dst, fanout, seed, _vp(out_src),
+                                       _vp(out_cnt))
 
     def items_cache_clear(self):
         self._lib.nts_items_cache_clear(self.h)
