@@ -142,13 +142,20 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                          "binary, resolved relative to the cfg) as the "
                          "workload instead of --graph")
     ap.add_argument("--model", default="gcn",
-                    choices=["gcn", "gat", "gcn-layer"],
+                    choices=["gcn", "gat", "gcn-layer", "gcn-sample"],
                     help="gcn = fused norm-degree aggregation (configs #2-4);"
                          " gat = attention-weighted layer with edge softmax "
                          "(config #5, single GPU, --feat 128); gcn-layer = "
                          "whole GCN layer with the feature-projection GEMM "
                          "ordered BEFORE the aggregation (SURVEY 8f-2: one "
-                         "HBM pass at f_out instead of f_in when f_out<<f_in)")
+                         "HBM pass at f_out instead of f_in when f_out<<f_in);"
+                         " gcn-sample = mini-batch step: GPU-resident "
+                         "reservoir sampling + 2-layer sampled aggregation "
+                         "(the reference's GCN_CPU_SAMPLE workload, 8f-3)")
+    ap.add_argument("--batch-size", type=int, default=4096,
+                    help="targets per step for --model gcn-sample")
+    ap.add_argument("--fanout", default="25,10",
+                    help="per-layer fan-outs for --model gcn-sample")
     ap.add_argument("--feat-out", type=int, default=128,
                     help="projection width for --model gcn-layer")
     ap.add_argument("--graph", default="reddit",
@@ -276,7 +283,53 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
     gy = torch.from_numpy(
         rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
 
-    if args.model == "gcn-layer":
+    if args.model == "gcn-sample":
+        # mini-batch step: sample a 2-layer subgraph on device
+        # (reservoir kernel + torch compaction), then aggregate innermost ->
+        # outermost with the same gather kernels.  Value still counts
+        # aggregated edges (the sampled edges of both layers, fwd+bwd).
+        assert not distributed, "sampled bench is single-GPU here"
+        from neutronstarlite_amd.ops import MiniBatchFuseOp, _u32_cuda
+        from neutronstarlite_amd.sampler_gpu import sample_subgraph_gpu
+        fanouts = [int(t) for t in args.fanout.split(",") if t]
+        ch0 = chunks[0]
+        d_coff = _u32_cuda(ch0.column_offset, dev)
+        d_rows = _u32_cuda(ch0.row_indices, dev)
+        import neutronstarlite_amd.graph as _G
+        outd_np, ind_np = _G.degrees(
+            np.stack([ch0.row_indices,
+                      np.repeat(np.arange(v, dtype=np.uint32),
+                                np.diff(ch0.column_offset.astype(np.int64)))],
+                     axis=1), v)
+        d_outd = torch.from_numpy(outd_np.astype(np.int64)).to(dev)
+        d_ind = torch.from_numpy(ind_np.astype(np.int64)).to(dev)
+        sampled_edges = [0]
+        step_idx = [0]
+        gen = np.random.default_rng(42)
+        target_pool = gen.permutation(v).astype(np.int32)
+
+        def step():
+            i = step_idx[0]
+            step_idx[0] += 1
+            lo_t = (i * args.batch_size) % max(1, v - args.batch_size)
+            targets = torch.from_numpy(
+                target_pool[lo_t:lo_t + args.batch_size]).to(dev)
+            layers = sample_subgraph_gpu(engine.stream, d_coff, d_rows,
+                                         targets, fanouts, d_outd, d_ind,
+                                         seed=1000 + i)
+            h = torch.randn(layers[-1].n_src, f, device=dev)
+            for ly in reversed(layers):
+                op = MiniBatchFuseOp(ly, dev, engine)
+                h = op.forward(h.contiguous())
+                sampled_edges[0] += ly.e_size
+            gy = torch.randn(layers[0].n_dst, f, device=dev)
+            g = gy
+            for ly in layers:
+                op = MiniBatchFuseOp(ly, dev, engine)
+                g = op.backward(g.contiguous())
+                sampled_edges[0] += ly.e_size
+            return h, g
+    elif args.model == "gcn-layer":
         # the layer's dense projection (x·W, the reference's
         # P[layer]->forward at NtsScheduler.hpp:737-740) ordered BEFORE the
         # aggregation: for 602->128 the gather streams 4.7x fewer bytes.
@@ -378,7 +431,13 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         log("cpu baseline (oracle, bounded sample)")
         cpu_baseline = cpu_baseline_leg(chunks, f)
 
-    value = args.steps * 2.0 * e_total / elapsed  # fwd + bwd edges, all ranks
+    if args.model == "gcn-sample":
+        # count the edges actually sampled+aggregated during the timed steps
+        # (warmup's share removed via the step counter)
+        value = sampled_edges[0] * (args.steps /
+                                    max(1, step_idx[0])) / elapsed
+    else:
+        value = args.steps * 2.0 * e_total / elapsed  # fwd+bwd, all ranks
     if rank == 0:
         out = {
             "metric": "aggregated_edges_per_sec",
@@ -403,6 +462,10 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                        f"GCN layer project({f}->{args.feat_out}) then "
                        "aggregate, fwd+bwd (SURVEY 8f-2 fused-layer order)"
                        if args.model == "gcn-layer" else
+                       f"mini-batch GCN: GPU-resident sampling batch="
+                       f"{args.batch_size} fanout={args.fanout} + sampled "
+                       "aggregation fwd+bwd (SURVEY 8f-3)"
+                       if args.model == "gcn-sample" else
                        "GCN-layer aggregation fwd+bwd"
                        + (" (BASELINE config #2)" if args.graph == "reddit"
                           else ""))),
