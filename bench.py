@@ -474,7 +474,7 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                 "parallelism": f"graph-partitioned dp{n}",
                 "edges_per_step": 2 * e_total,
             },
-            "roofline": roofline,
+            "roofline": (None if args.model == "gcn-sample" else roofline),
             "cpu_baseline": cpu_baseline,
         }
         print(json.dumps(out), flush=True)
