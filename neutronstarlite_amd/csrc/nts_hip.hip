@@ -570,6 +570,43 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
 /* per-edge dot: out[e] = dot(dst_rows[d], src_rows[row_indices[e]-src_s]),
  * item-driven (bounded per-wave work even on power-law hubs); lanes stride
  * the feature dim per edge. */
+/* lane-per-edge variant of the edge dot: each lane owns one edge and walks
+ * its source row sequentially (per-lane streaming; L1 serves the row's
+ * cache lines) — 64 independent dot products in flight per wave vs one for
+ * the wave-per-edge form.  Default; NTS_EDGE_DOT=1 selects wave-per-edge. */
+__global__ void k_edge_dot_lpe(const uint4 *__restrict__ items,
+                               const uint32_t *__restrict__ n_items_p,
+                               float *__restrict__ out,
+                               const float *__restrict__ dst_rows,
+                               const float *__restrict__ src_rows,
+                               const uint32_t *__restrict__ row_indices,
+                               uint32_t src_start, uint32_t f) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y, cnt = itm.z;
+    const float *a = dst_rows + (uint64_t)d * f;
+    for (uint32_t ei = lane; ei < cnt; ei += 64) {
+      const uint32_t e = e0 + ei;
+      const float *b = src_rows + (uint64_t)(row_indices[e] - src_start) * f;
+      float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+      uint32_t j = 0;
+      for (; j + 4 <= f; j += 4) {
+        s0 = fmaf(a[j], b[j], s0);
+        s1 = fmaf(a[j + 1], b[j + 1], s1);
+        s2 = fmaf(a[j + 2], b[j + 2], s2);
+        s3 = fmaf(a[j + 3], b[j + 3], s3);
+      }
+      for (; j < f; ++j) s0 = fmaf(a[j], b[j], s0);
+      out[e] = ((s0 + s1) + (s2 + s3));
+    }
+  }
+}
+
 __global__ void k_edge_dot(const uint4 *__restrict__ items,
                            const uint32_t *__restrict__ n_items_p,
                            float *__restrict__ out,
@@ -1028,9 +1065,16 @@ void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
   Tic t(s, NTS_KTAG_EDGE);
   const uint32_t grid =
       grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
-  hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
-                     ib.items, ib.counter, out, dst_rows, src_rows,
-                     row_indices, src_start, feature_size);
+  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 2);
+  if (variant == 1) {
+    hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
+                       ib.items, ib.counter, out, dst_rows, src_rows,
+                       row_indices, src_start, feature_size);
+  } else {
+    hipLaunchKernelGGL(k_edge_dot_lpe, dim3(grid), dim3(NTS_BLOCK), 0,
+                       s->stream, ib.items, ib.counter, out, dst_rows,
+                       src_rows, row_indices, src_start, feature_size);
+  }
   dbg_sync(s, "k_edge_dot");
 }
 
