@@ -573,7 +573,9 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
 /* lane-per-edge variant of the edge dot: each lane owns one edge and walks
  * its source row sequentially (per-lane streaming; L1 serves the row's
  * cache lines) — 64 independent dot products in flight per wave vs one for
- * the wave-per-edge form.  Default; NTS_EDGE_DOT=1 selects wave-per-edge. */
+ * the wave-per-edge form.  Measured SLOWER (42.4 vs 34.3 ms/step on
+ * config #5): kept selectable via NTS_EDGE_DOT=2 as a recorded
+ * negative result. */
 __global__ void k_edge_dot_lpe(const uint4 *__restrict__ items,
                                const uint32_t *__restrict__ n_items_p,
                                float *__restrict__ out,
@@ -1065,7 +1067,10 @@ void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
   Tic t(s, NTS_KTAG_EDGE);
   const uint32_t grid =
       grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
-  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 2);
+  /* measured on config #5: wave-per-edge 34.3 ms/step vs
+   * lane-per-edge 42.4 — the per-lane scattered rows thrash L1
+   * harder than the serial-but-coalesced row reads win back */
+  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 1);
   if (variant == 1) {
     hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
                        ib.items, ib.counter, out, dst_rows, src_rows,
