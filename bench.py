@@ -142,7 +142,8 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                          "binary, resolved relative to the cfg) as the "
                          "workload instead of --graph")
     ap.add_argument("--model", default="gcn",
-                    choices=["gcn", "gat", "gcn-layer", "gcn-sample"],
+                    choices=["gcn", "gat", "gcn-layer", "gcn-sample",
+                             "gcn-train"],
                     help="gcn = fused norm-degree aggregation (configs #2-4);"
                          " gat = attention-weighted layer with edge softmax "
                          "(config #5, single GPU, --feat 128); gcn-layer = "
@@ -151,7 +152,10 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                          "HBM pass at f_out instead of f_in when f_out<<f_in);"
                          " gcn-sample = mini-batch step: GPU-resident "
                          "reservoir sampling + 2-layer sampled aggregation "
-                         "(the reference's GCN_CPU_SAMPLE workload, 8f-3)")
+                         "(the reference's GCN_CPU_SAMPLE workload, 8f-3);"
+                         " gcn-train = full 2-layer GCN training epoch "
+                         "(agg+mm+relu+agg+mm+nll fwd/bwd + Adam, the "
+                         "ALGORITHM:GCN train loop, GCN.hpp:237-306)")
     ap.add_argument("--batch-size", type=int, default=4096,
                     help="targets per step for --model gcn-sample")
     ap.add_argument("--fanout", default="25,10",
@@ -283,7 +287,33 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
     gy = torch.from_numpy(
         rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
 
-    if args.model == "gcn-sample":
+    if args.model == "gcn-train":
+        # end-to-end training epoch like GCN_impl::run (toolkits/GCN.hpp:237):
+        # layer0 aggregate(f=602) -> W0 -> relu -> layer1 aggregate(128) ->
+        # W1 -> log_softmax -> nll; backward through the autograd-bridged
+        # aggregation; Adam step.  A "step" here is one full epoch.
+        assert not distributed, "training bench is single-GPU here"
+        from neutronstarlite_amd.ops import aggregate
+        f1, ncls = args.feat_out, 41   # LAYERS 602-128-41 (gcn_reddit.cfg)
+        dch = dchunks[0]
+        gw = torch.Generator(device="cpu").manual_seed(7)
+        W0 = (torch.rand(f, f1, generator=gw) * 0.2 - 0.1).to(dev).requires_grad_(True)
+        W1 = (torch.rand(f1, ncls, generator=gw) * 0.2 - 0.1).to(dev).requires_grad_(True)
+        labels = torch.randint(ncls, (hi - lo,),
+                               generator=gw).to(dev)
+        opt = torch.optim.Adam([W0, W1], lr=1e-2)
+
+        def step():
+            opt.zero_grad(set_to_none=True)
+            a0 = aggregate(x, dch, engine)
+            h0 = torch.relu(a0 @ W0)
+            a1 = aggregate(h0, dch, engine)
+            out = torch.log_softmax(a1 @ W1, 1)
+            loss = torch.nn.functional.nll_loss(out, labels)
+            loss.backward()
+            opt.step()
+            return loss
+    elif args.model == "gcn-sample":
         # mini-batch step: sample a 2-layer subgraph on device
         # (reservoir kernel + torch compaction), then aggregate innermost ->
         # outermost with the same gather kernels.  Value still counts
@@ -431,7 +461,33 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         log("cpu baseline (oracle, bounded sample)")
         cpu_baseline = cpu_baseline_leg(chunks, f)
 
-    if args.model == "gcn-sample":
+    if args.model == "gcn-train":
+        # end-to-end training epoch like GCN_impl::run (toolkits/GCN.hpp:237):
+        # layer0 aggregate(f=602) -> W0 -> relu -> layer1 aggregate(128) ->
+        # W1 -> log_softmax -> nll; backward through the autograd-bridged
+        # aggregation; Adam step.  A "step" here is one full epoch.
+        assert not distributed, "training bench is single-GPU here"
+        from neutronstarlite_amd.ops import aggregate
+        f1, ncls = args.feat_out, 41   # LAYERS 602-128-41 (gcn_reddit.cfg)
+        dch = dchunks[0]
+        gw = torch.Generator(device="cpu").manual_seed(7)
+        W0 = (torch.rand(f, f1, generator=gw) * 0.2 - 0.1).to(dev).requires_grad_(True)
+        W1 = (torch.rand(f1, ncls, generator=gw) * 0.2 - 0.1).to(dev).requires_grad_(True)
+        labels = torch.randint(ncls, (hi - lo,),
+                               generator=gw).to(dev)
+        opt = torch.optim.Adam([W0, W1], lr=1e-2)
+
+        def step():
+            opt.zero_grad(set_to_none=True)
+            a0 = aggregate(x, dch, engine)
+            h0 = torch.relu(a0 @ W0)
+            a1 = aggregate(h0, dch, engine)
+            out = torch.log_softmax(a1 @ W1, 1)
+            loss = torch.nn.functional.nll_loss(out, labels)
+            loss.backward()
+            opt.step()
+            return loss
+    elif args.model == "gcn-sample":
         # count the edges actually sampled+aggregated during the timed steps
         # (warmup's share removed via the step counter)
         value = sampled_edges[0] * (args.steps /
@@ -462,6 +518,9 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                        f"GCN layer project({f}->{args.feat_out}) then "
                        "aggregate, fwd+bwd (SURVEY 8f-2 fused-layer order)"
                        if args.model == "gcn-layer" else
+                       f"2-layer GCN training epoch ({f}-{args.feat_out}-41,"
+                       " agg+mm fwd/bwd + Adam; ALGORITHM:GCN loop)"
+                       if args.model == "gcn-train" else
                        f"mini-batch GCN: GPU-resident sampling batch="
                        f"{args.batch_size} fanout={args.fanout} + sampled "
                        "aggregation fwd+bwd (SURVEY 8f-3)"

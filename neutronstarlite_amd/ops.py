@@ -89,6 +89,35 @@ class HipEngine:
             ch.edge_size, ch.src_n, f, with_weight)
 
 
+class _AggregateFn(torch.autograd.Function):
+    """torch.autograd bridge over the fused aggregation: forward = CSC pull,
+    backward = CSR push (the ntsGraphOp tape contract of
+    ntsContext::self_backward, ntsContext.hpp:276-359, expressed as a torch
+    Function so whole training loops are plain torch code)."""
+
+    @staticmethod
+    def forward(ctx, x, dchunk, engine):
+        ctx.dchunk = dchunk
+        ctx.engine = engine
+        y = torch.zeros(dchunk.dst_n, x.shape[1], dtype=torch.float32,
+                        device=x.device)
+        engine.csc_forward(dchunk, x.contiguous(), y)
+        return y
+
+    @staticmethod
+    def backward(ctx, grad_y):
+        gx = torch.zeros(ctx.dchunk.src_n, grad_y.shape[1],
+                         dtype=torch.float32, device=grad_y.device)
+        ctx.engine.csr_backward(ctx.dchunk, grad_y.contiguous(), gx)
+        return gx, None, None
+
+
+def aggregate(x: torch.Tensor, dchunk: "DeviceChunk",
+              engine: "HipEngine") -> torch.Tensor:
+    """Autograd-aware fused aggregation (single-GPU chunk)."""
+    return _AggregateFn.apply(x, dchunk, engine)
+
+
 class MiniBatchFuseOp:
     """MiniBatchFuseOp equivalent (core/ntsMiniBatchGraphOp.hpp:61-131):
     the same aggregation arithmetic on one sampled layer's compacted
