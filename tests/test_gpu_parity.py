@@ -18,7 +18,7 @@ RTOL, ATOL = 1e-4, 1e-5
 
 
 def assert_close(got, ref, name=""):
-    got = got.cpu().numpy() if isinstance(got, torch.Tensor) else got
+    got = got.detach().cpu().numpy() if isinstance(got, torch.Tensor) else got
     err = np.abs(got - ref)
     tol = RTOL * np.abs(ref) + ATOL
     bad = err > tol
