@@ -462,31 +462,8 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         cpu_baseline = cpu_baseline_leg(chunks, f)
 
     if args.model == "gcn-train":
-        # end-to-end training epoch like GCN_impl::run (toolkits/GCN.hpp:237):
-        # layer0 aggregate(f=602) -> W0 -> relu -> layer1 aggregate(128) ->
-        # W1 -> log_softmax -> nll; backward through the autograd-bridged
-        # aggregation; Adam step.  A "step" here is one full epoch.
-        assert not distributed, "training bench is single-GPU here"
-        from neutronstarlite_amd.ops import aggregate
-        f1, ncls = args.feat_out, 41   # LAYERS 602-128-41 (gcn_reddit.cfg)
-        dch = dchunks[0]
-        gw = torch.Generator(device="cpu").manual_seed(7)
-        W0 = (torch.rand(f, f1, generator=gw) * 0.2 - 0.1).to(dev).requires_grad_(True)
-        W1 = (torch.rand(f1, ncls, generator=gw) * 0.2 - 0.1).to(dev).requires_grad_(True)
-        labels = torch.randint(ncls, (hi - lo,),
-                               generator=gw).to(dev)
-        opt = torch.optim.Adam([W0, W1], lr=1e-2)
-
-        def step():
-            opt.zero_grad(set_to_none=True)
-            a0 = aggregate(x, dch, engine)
-            h0 = torch.relu(a0 @ W0)
-            a1 = aggregate(h0, dch, engine)
-            out = torch.log_softmax(a1 @ W1, 1)
-            loss = torch.nn.functional.nll_loss(out, labels)
-            loss.backward()
-            opt.step()
-            return loss
+        # one epoch = 2 layers x (fwd + bwd) aggregation
+        value = args.steps * 4.0 * e_total / elapsed
     elif args.model == "gcn-sample":
         # count the edges actually sampled+aggregated during the timed steps
         # (warmup's share removed via the step counter)
