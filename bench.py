@@ -302,10 +302,17 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         labels = torch.randint(ncls, (hi - lo,),
                                generator=gw).to(dev)
         opt = torch.optim.Adam([W0, W1], lr=1e-2)
+        # the reference's tape always runs every graph op's backward
+        # (ntsContext::self_backward, ntsContext.hpp:276-359) — require grad
+        # on the features so layer 0's backward aggregation runs here too
+        # and the counted 2x(fwd+bwd) passes are all real
+        x_t = x.clone().requires_grad_(True)
 
         def step():
             opt.zero_grad(set_to_none=True)
-            a0 = aggregate(x, dch, engine)
+            if x_t.grad is not None:
+                x_t.grad = None
+            a0 = aggregate(x_t, dch, engine)
             h0 = torch.relu(a0 @ W0)
             a1 = aggregate(h0, dch, engine)
             out = torch.log_softmax(a1 @ W1, 1)
