@@ -205,6 +205,12 @@ void nts_sample_reservoir(nts_stream *s, const nts_vid *column_offset,
     nts_vid fanout, unsigned long long seed, nts_vid *out_src,
     nts_vid *out_cnt);
 
+/* gather-permute (additive): out[i] = in[index[i]] for f32 values and u32
+ * indices — carries per-edge values between CSC and CSR edge order (e.g.
+ * attention weights for the backward gather) without host round trips. */
+void nts_permute_f32(nts_stream *s, float *out, const float *in,
+    const nts_vid *index, long n);
+
 /* Device info for the host layer / bench. */
 int nts_device_count(void);
 void nts_set_device(int dev);

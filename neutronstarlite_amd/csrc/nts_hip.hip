@@ -446,6 +446,18 @@ __global__ void k_agg_msg(float *__restrict__ master,
   }
 }
 
+/* gather-permute: out[i] = in[index[i]] (f32, u32 indices) — carries
+ * per-edge values between CSC and CSR edge order without torch's int64
+ * index machinery */
+__global__ void k_permute_f32(float *__restrict__ out,
+                              const float *__restrict__ in,
+                              const uint32_t *__restrict__ index,
+                              uint64_t n) {
+  for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x)
+    out[i] = in[index[i]];
+}
+
 /* dense-row pack/unpack for the RCCL ring (indices static, payload dense) */
 enum RowOp { ROW_GATHER, ROW_SCATTER, ROW_SCATTER_ADD };
 template <RowOp OP>
@@ -605,6 +617,38 @@ __global__ void k_edge_dot_lpe(const uint4 *__restrict__ items,
       }
       for (; j < f; ++j) s0 = fmaf(a[j], b[j], s0);
       out[e] = ((s0 + s1) + (s2 + s3));
+    }
+  }
+}
+
+template <int GS>  /* lanes per edge; 64/GS edges concurrent per wave */
+__global__ void k_edge_dot_sg(const uint4 *__restrict__ items,
+                              const uint32_t *__restrict__ n_items_p,
+                              float *__restrict__ out,
+                              const float *__restrict__ dst_rows,
+                              const float *__restrict__ src_rows,
+                              const uint32_t *__restrict__ row_indices,
+                              uint32_t src_start, uint32_t f) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  const uint32_t sub = lane / GS;           /* which edge of the pair/quad */
+  const uint32_t sl = lane % GS;
+  constexpr uint32_t NSUB = 64 / GS;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y, cnt = itm.z;
+    const float *a = dst_rows + (uint64_t)d * f;
+    for (uint32_t k = sub; k < cnt; k += NSUB) {
+      const uint32_t e = e0 + k;
+      const float *b = src_rows + (uint64_t)(row_indices[e] - src_start) * f;
+      float sum = 0.f;
+      for (uint32_t j = sl; j < f; j += GS) sum += a[j] * b[j];
+#pragma unroll
+      for (int w = GS / 2; w >= 1; w >>= 1) sum += __shfl_xor(sum, w, GS);
+      if (sl == 0) out[e] = sum;
     }
   }
 }
@@ -1067,18 +1111,31 @@ void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
   Tic t(s, NTS_KTAG_EDGE);
   const uint32_t grid =
       grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
-  /* measured on config #5: wave-per-edge 34.3 ms/step vs
-   * lane-per-edge 42.4 — the per-lane scattered rows thrash L1
-   * harder than the serial-but-coalesced row reads win back */
-  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 1);
+  /* measured on config #5 (f=128): wave-per-edge 34.3 ms/step beats
+   * lane-per-edge 42.4 (scattered rows thrash L1); the sub-group form
+   * (GS lanes per edge, 64/GS edges in flight) is the default.
+   * NTS_EDGE_DOT: 1 = wave-per-edge, 2 = lane-per-edge, 3 = sub-group. */
+  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 3);
   if (variant == 1) {
     hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
                        ib.items, ib.counter, out, dst_rows, src_rows,
                        row_indices, src_start, feature_size);
-  } else {
+  } else if (variant == 2) {
     hipLaunchKernelGGL(k_edge_dot_lpe, dim3(grid), dim3(NTS_BLOCK), 0,
                        s->stream, ib.items, ib.counter, out, dst_rows,
                        src_rows, row_indices, src_start, feature_size);
+  } else if (feature_size <= 32) {
+    hipLaunchKernelGGL(k_edge_dot_sg<16>, dim3(grid), dim3(NTS_BLOCK), 0,
+                       s->stream, ib.items, ib.counter, out, dst_rows,
+                       src_rows, row_indices, src_start, feature_size);
+  } else if (feature_size <= 256) {
+    hipLaunchKernelGGL(k_edge_dot_sg<32>, dim3(grid), dim3(NTS_BLOCK), 0,
+                       s->stream, ib.items, ib.counter, out, dst_rows,
+                       src_rows, row_indices, src_start, feature_size);
+  } else {
+    hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
+                       ib.items, ib.counter, out, dst_rows, src_rows,
+                       row_indices, src_start, feature_size);
   }
   dbg_sync(s, "k_edge_dot");
 }
@@ -1141,6 +1198,14 @@ void nts_sample_reservoir(nts_stream *s, const nts_vid *column_offset,
                      s->stream, column_offset, row_indices, dst_list, n_dst,
                      fanout, seed, out_src, out_cnt);
   dbg_sync(s, "k_sample_reservoir");
+}
+
+void nts_permute_f32(nts_stream *s, float *out, const float *in,
+                     const nts_vid *index, long n) {
+  if (n <= 0) return;
+  hipLaunchKernelGGL(k_permute_f32, dim3(grid_for((uint64_t)n)),
+                     dim3(NTS_BLOCK), 0, s->stream, out, in, index,
+                     (uint64_t)n);
 }
 
 int nts_device_count(void) {

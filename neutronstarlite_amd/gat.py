@@ -46,6 +46,8 @@ class GATLayer:
         # as the stable-by-src permutation of the CSC order).
         perm = np.argsort(self._src_of_edge_np, kind="stable")
         self.csr_from_csc = torch.from_numpy(perm).to(device)
+        self._perm_u32 = torch.from_numpy(
+            perm.astype(np.uint32).view(np.int32)).to(device)
         self._doe = self._soe = None
         self._ones_dst = self._ones_src = None
         # separate wrapper of the same HIP stream for the scalar (f=1)
@@ -105,7 +107,10 @@ class GATLayer:
         f = grad_y.shape[1]
         grad_y = grad_y.contiguous()
         # CSR gather with attention weights permuted into CSR edge order
-        s_bwd = saved["s"][self.csr_from_csc].contiguous()
+        # (nts_permute_f32: ~8x torch's int64 index_select at this size)
+        s_bwd = torch.empty(E, 1, device=dev)
+        st.permute_f32(s_bwd.data_ptr(), saved["s"].data_ptr(),
+                       self._perm_u32.data_ptr(), E)
         grad_h = torch.zeros(ch.src_n, f, device=dev)
         st.gather_by_src_from_dst(grad_y.data_ptr(), grad_h.data_ptr(),
                                   s_bwd.data_ptr(),
@@ -135,7 +140,9 @@ class GATLayer:
         if self._ones_dst is None:
             self._ones_dst = torch.ones(ch.dst_n, 1, device=dev)
             self._ones_src = torch.ones(ch.src_n, 1, device=dev)
-        ge_csr = ge[self.csr_from_csc].contiguous()
+        ge_csr = torch.empty(E, 1, device=dev)
+        st.permute_f32(ge_csr.data_ptr(), ge.data_ptr(),
+                       self._perm_u32.data_ptr(), E)
         sst = self.scalar_stream
         g_src = torch.zeros(ch.src_n, 1, device=dev)
         sst.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),

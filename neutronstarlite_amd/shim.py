@@ -87,6 +87,7 @@ def lib():
     l.nts_edge_dot.argtypes = [_vp] + [_vp] * 5 + [_u32] * 3
     l.nts_edge_softmax_forward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_permute_f32.argtypes = [_vp, _vp, _vp, _vp, _i64]
     l.nts_sample_reservoir.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32,
                                        _c.c_ulonglong, _vp, _vp]
     l.nts_device_count.restype = _i32
@@ -238,6 +239,9 @@ class Stream:
         self._lib.nts_edge_softmax_backward(
             self.h, _vp(msg_in_grad), _vp(msg_out_grad), _vp(msg_cached),
             _vp(row_indices), _vp(column_offset), batch, f)
+
+    def permute_f32(self, out, inp, index, n):
+        self._lib.nts_permute_f32(self.h, _vp(out), _vp(inp), _vp(index), n)
 
     def sample_reservoir(self, column_offset, row_indices, dst_list, n_dst,
                          fanout, seed, out_src, out_cnt):
