@@ -1113,9 +1113,10 @@ void nts_edge_dot(nts_stream *s, float *out, const float *dst_rows,
       grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
   /* measured on config #5 (f=128): wave-per-edge 34.3 ms/step beats
    * lane-per-edge 42.4 (scattered rows thrash L1); the sub-group form
-   * (GS lanes per edge, 64/GS edges in flight) is the default.
-   * NTS_EDGE_DOT: 1 = wave-per-edge, 2 = lane-per-edge, 3 = sub-group. */
-  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 3);
+   * (GS lanes per edge) also measured worse at f=128 (~+6 ms/step).
+   * NTS_EDGE_DOT: 1 = wave-per-edge (default), 2 = lane-per-edge,
+   * 3 = sub-group — both alternatives kept as recorded negative results. */
+  static const uint32_t variant = env_u32("NTS_EDGE_DOT", 1);
   if (variant == 1) {
     hipLaunchKernelGGL(k_edge_dot, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
                        ib.items, ib.counter, out, dst_rows, src_rows,
