@@ -2,8 +2,6 @@
 generating script (which reads /root/reference) must reproduce the committed
 files bit-exactly.  Skipped where the reference is not mounted (GPU boxes)."""
 import os
-import runpy
-import shutil
 
 import numpy as np
 import pytest
