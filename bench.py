@@ -282,8 +282,23 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         setup_mirror_lists(rg)
 
     rng = np.random.default_rng(42 + rank)
-    x = torch.from_numpy(
-        rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
+    x_np = None
+    if args.cfg:
+        # use the cfg's FEATURE_FILE when it exists (GNNDatum text format);
+        # otherwise fall back to the synthetic convention
+        from neutronstarlite_amd.config import read_cfg
+        from neutronstarlite_amd import data as D
+        info = read_cfg(args.cfg)
+        fpath = info.feature_file
+        if fpath and not os.path.isabs(fpath):
+            fpath = os.path.join(os.path.dirname(os.path.abspath(args.cfg)),
+                                 fpath)
+        if fpath and os.path.exists(fpath):
+            x_np = D.read_feature_table(fpath, lo, hi, f)
+            log(f"features from {fpath}")
+    if x_np is None:
+        x_np = rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)
+    x = torch.from_numpy(x_np).to(dev)
     gy = torch.from_numpy(
         rng.uniform(-1, 1, size=(hi - lo, f)).astype(np.float32)).to(dev)
 
