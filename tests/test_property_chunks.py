@@ -84,3 +84,26 @@ def test_oracle_forward_equals_dense_matmul(data):
     A = np.zeros((v, v))
     np.add.at(A, (edges[:, 1], edges[:, 0]), w.astype(np.float64))
     assert np.allclose(y, A @ x, rtol=1e-4, atol=1e-5)
+
+
+@given(edge_lists(), st.integers(1, 12))
+@settings(max_examples=25, deadline=None)
+def test_host_sampler_contract(data, fanout):
+    """Host sampler: <= fanout slots per destination, all real edge slots,
+    for arbitrary graphs (multigraphs included)."""
+    from neutronstarlite_amd.sampler import sample_layer
+    v, edges = data
+    if len(edges) == 0:
+        return
+    outd, ind = G.degrees(edges, v)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    ch = G.build_chunks(edges, w, np.array([0, v], dtype=np.uint32), 0)[0]
+    rng = np.random.default_rng(0)
+    dst = np.unique(rng.integers(0, v, size=min(v, 16)).astype(np.uint32))
+    ly = sample_layer(ch.column_offset, ch.row_indices, dst, fanout,
+                      outd, ind, rng)
+    deg_full = (ch.column_offset[dst + 1]
+                - ch.column_offset[dst]).astype(np.int64)
+    deg_s = np.diff(ly.column_offset.astype(np.int64))
+    assert np.array_equal(deg_s, np.minimum(deg_full, fanout))
+    assert np.array_equal(ly.src[ly.row_indices_local], ly.row_indices_global)
