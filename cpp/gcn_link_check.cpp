@@ -1,0 +1,185 @@
+/* gcn_link_check — proves the north_star link contract for the flagship
+ * dist-GCN toolkit: the per-layer loop of toolkits/GCN.hpp:217-235
+ * (runGraphOp<nts::op::ForwardGPUfuseOp> + two-input runVertexForward)
+ * compiles against cpp/include/nts/nts.hpp with the reference's names and
+ * signatures.  Compiling and linking this TU IS the contract test (run by
+ * __graft_entry__.build() on every round); executing it with a GPU
+ * additionally runs the loop at world-1 RCCL — ncclCommInitAll +
+ * grouped self send/recv + allreduce + broadcast all execute on hardware
+ *  — and checks the P=1 ForwardGPUfuseOp output
+ * bit-matches ForwardSingleGPUfuseOp.
+ *
+ * Loop shape vendored (structure, not code) from toolkits/GCN.hpp:
+ *   Forward()            GCN.hpp:217-235
+ *   vertexForward(a, x)  GCN.hpp:183-195 (relu/log_softmax of P[l]->forward)
+ *   Update() allreduce   GCN.hpp:211-214 -> Network_simple::all_reduce_sum
+ *                        (comm/network.h:198-203) -> nts_comm_allreduce_sum_f32
+ *
+ * Usage: gcn_link_check    (exit 0 always when the binary exists; GPU parts
+ *                           run only if torch::cuda::is_available())
+ */
+#include <torch/torch.h>
+
+#include <cmath>
+#include <cstdio>
+#include <vector>
+
+#include "nts/nts.hpp"
+
+using namespace nts;
+
+/* ---- the vendored loop shape (GCN.hpp:217-235) ---- */
+struct GCNLoop {
+  PartitionedGraph *partitioned_graph;
+  VertexSubset *active;
+  NtsContext *ctx;
+  std::vector<NtsVar> X; /* X[0..layers] as in GCN.hpp:108-117 */
+  std::vector<NtsVar> W;
+
+  NtsVar vertexForward(NtsVar &a, NtsVar &x, int layer) {
+    (void)x; /* GCN.hpp:183-195: second input unused by the GCN model */
+    if (layer == 0) return torch::relu(torch::mm(a, W[0]));
+    return torch::log_softmax(torch::mm(a, W[1]), 1);
+  }
+
+  void Forward(int layers) {
+    for (int i = 0; i < layers; i++) {
+      /* GCN.hpp:227-232, names unchanged */
+      NtsVar Y_i = ctx->runGraphOp<nts::op::ForwardGPUfuseOp>(
+          partitioned_graph, active, X[i]);
+      X[i + 1] = ctx->runVertexForward(
+          [&](NtsVar &n_i, NtsVar &v_i) {
+            return vertexForward(n_i, v_i, i);
+          },
+          Y_i, X[i]);
+    }
+  }
+};
+
+int main() {
+  if (!torch::cuda::is_available()) {
+    /* compile+link already proved the contract */
+    printf("gcn_link_check: link ok (no GPU; runtime part skipped)\n");
+    return 0;
+  }
+  const VertexId V = 2048;
+  const uint32_t E = 20000;
+  const int F0 = 32, F1 = 16, C = 5;
+  torch::manual_seed(3);
+  torch::Device dev(torch::kCUDA, 0);
+
+  /* small synthetic graph with self loops */
+  std::vector<uint32_t> src, dst;
+  {
+    auto s = torch::randint(V, {E}, torch::kLong);
+    auto d = torch::randint(V, {E}, torch::kLong);
+    for (uint32_t i = 0; i < E; i++) {
+      src.push_back((uint32_t)s[i].item<int64_t>());
+      dst.push_back((uint32_t)d[i].item<int64_t>());
+    }
+    for (VertexId v = 0; v < V; v++) { src.push_back(v); dst.push_back(v); }
+  }
+  const uint32_t Etot = src.size();
+  std::vector<uint32_t> outd(V, 0), ind(V, 0);
+  for (uint32_t e = 0; e < Etot; e++) { outd[src[e]]++; ind[dst[e]]++; }
+  for (VertexId v = 0; v < V; v++) { if (!outd[v]) outd[v] = 1; if (!ind[v]) ind[v] = 1; }
+  std::vector<uint32_t> col_off(V + 1, 0), row_off(V + 1, 0);
+  for (uint32_t e = 0; e < Etot; e++) { col_off[dst[e] + 1]++; row_off[src[e] + 1]++; }
+  for (VertexId v = 0; v < V; v++) { col_off[v + 1] += col_off[v]; row_off[v + 1] += row_off[v]; }
+  std::vector<uint32_t> rows(Etot), cols(Etot), cpos = col_off, rpos = row_off;
+  std::vector<float> wf(Etot), wb(Etot);
+  for (uint32_t e = 0; e < Etot; e++) {
+    float w = 1.0f / (std::sqrt((float)outd[src[e]]) * std::sqrt((float)ind[dst[e]]));
+    uint32_t pc = cpos[dst[e]]++, pr = rpos[src[e]]++;
+    rows[pc] = src[e]; wf[pc] = w;
+    cols[pr] = dst[e]; wb[pr] = w;
+  }
+  auto chunk = CSC_segment_pinned::from_host(
+      0, V, 0, V, col_off.data(), rows.data(), wf.data(), row_off.data(),
+      cols.data(), wb.data(), Etot, dev);
+
+  /* world-1 RCCL communicator: the ring degenerates but RCCL init and the
+   * collective entry points execute for real on this box */
+  nts_comm *comm = nullptr;
+  int dev0 = 0;
+  if (nts_comm_init_all(&comm, 1, &dev0) != 0 || !comm) {
+    fprintf(stderr, "nts_comm_init_all failed\n");
+    return 1;
+  }
+  printf("rccl: world=%d rank=%d\n", nts_comm_size(comm), nts_comm_rank(comm));
+
+  PartitionedGraph pg;
+  pg.graph_chunks.push_back(&chunk);
+  pg.partition_offset = {0, V};
+  pg.comm = comm;
+  VertexSubset active{0, V};
+
+  /* grouped self send/recv over RCCL (the ring-step primitive) */
+  {
+    NtsVar a = torch::rand({1024}, torch::device(dev));
+    NtsVar b = torch::zeros({1024}, torch::device(dev));
+    nts_comm_group_begin();
+    nts_comm_send_f32(comm, pg.stream, a.data_ptr<float>(), 1024, 0);
+    nts_comm_recv_f32(comm, pg.stream, b.data_ptr<float>(), 1024, 0);
+    nts_comm_group_end();
+    nts_stream_sync(pg.stream);
+    if (!torch::equal(a, b)) { fprintf(stderr, "self sendrecv mismatch\n"); return 1; }
+    printf("rccl grouped self send/recv ok\n");
+  }
+
+  /* the flagship loop, world 1 */
+  GCNLoop loop;
+  loop.partitioned_graph = &pg;
+  loop.active = &active;
+  NtsContext ctx;
+  loop.ctx = &ctx;
+  loop.X.resize(3);
+  loop.X[0] = torch::rand({(int64_t)V, F0}, torch::device(dev)) * 2 - 1;
+  loop.W.push_back(torch::empty({F0, F1}, torch::device(dev))
+                       .uniform_(-0.1, 0.1).set_requires_grad(true));
+  loop.W.push_back(torch::empty({F1, C}, torch::device(dev))
+                       .uniform_(-0.1, 0.1).set_requires_grad(true));
+  loop.Forward(2);
+  NtsVar labels = torch::randint(C, {(int64_t)V},
+                                 torch::device(dev).dtype(torch::kLong));
+  NtsVar loss = torch::nll_loss(loop.X[2], labels);
+  ctx.self_backward(loss);
+  printf("flagship loop fwd+bwd ok, loss %.4f\n", loss.item<double>());
+
+  /* weight-grad allreduce + init bcast (Update()'s comm, world 1) */
+  for (auto &Wl : loop.W) {
+    NtsVar g = Wl.grad();
+    if (!g.defined()) { fprintf(stderr, "missing weight grad\n"); return 1; }
+    NtsVar gc = g.contiguous();
+    nts_comm_allreduce_sum_f32(comm, pg.stream, gc.data_ptr<float>(),
+                               gc.data_ptr<float>(), gc.numel());
+    nts_comm_bcast_f32(comm, pg.stream, Wl.data_ptr<float>(), Wl.numel(), 0);
+  }
+  nts_stream_sync(pg.stream);
+  printf("rccl allreduce+bcast ok\n");
+
+  /* P=1 equivalence: ForwardGPUfuseOp must match ForwardSingleGPUfuseOp */
+  {
+    NtsVar x = torch::rand({(int64_t)V, 24}, torch::device(dev)) * 2 - 1;
+    op::ForwardGPUfuseOp dist(&pg, &active);
+    op::ForwardSingleGPUfuseOp single(&pg, &active);
+    NtsVar yd = dist.forward(x);
+    NtsVar ys = single.forward(x);
+    if (!torch::equal(yd, ys)) {
+      fprintf(stderr, "P=1 ForwardGPUfuseOp != ForwardSingleGPUfuseOp\n");
+      return 1;
+    }
+    NtsVar g = torch::rand({(int64_t)V, 24}, torch::device(dev)) * 2 - 1;
+    NtsVar gd = dist.backward(g);
+    NtsVar gs = single.backward(g);
+    if (!torch::equal(gd, gs)) {
+      fprintf(stderr, "P=1 backward mismatch\n");
+      return 1;
+    }
+    printf("P=1 dist==single parity ok\n");
+  }
+
+  nts_comm_destroy(comm);
+  printf("gcn_link_check ok\n");
+  return 0;
+}

@@ -25,6 +25,7 @@
 #include <functional>
 #include <vector>
 
+#include "nts_comm.h"
 #include "nts_hip.h"
 
 namespace nts {
@@ -89,6 +90,10 @@ struct PartitionedGraph {
   int partition_id = 0;
   std::vector<VertexId> partition_offset;  // [P+1]
   nts_stream *stream = nullptr;            // compute stream (C-ABI)
+  nts_comm *comm = nullptr;                // RCCL ring (nullptr => P==1 only)
+  torch::Tensor mirror_index;              // per-graph mirror compression
+                                           // (generateMirrorIndex surface,
+                                           // PartitionedGraph.hpp:295-305)
 
   /* Wrap the HIP null stream (= torch's default stream): libtorch tensor
    * fills/copies and our kernels must share one stream order. */
@@ -98,6 +103,24 @@ struct PartitionedGraph {
   }
   VertexId owned_vertices() const {
     return partition_offset[partition_id + 1] - partition_offset[partition_id];
+  }
+  int partitions() const { return (int)partition_offset.size() - 1; }
+  VertexId part_n(int k) const {
+    return partition_offset[k + 1] - partition_offset[k];
+  }
+
+  /* Single-partition mirror index = identity over the whole graph (the
+   * multi-partition compressed index is built at partition time).  Stored
+   * per graph object — NOT function-static — so several graphs/devices can
+   * coexist in one process. */
+  torch::Tensor &mirror_index_for(torch::Device dev) {
+    const int64_t v = partition_offset.back();
+    if (!mirror_index.defined() || mirror_index.size(0) != v ||
+        mirror_index.device() != dev) {
+      mirror_index = torch::arange(
+          v, torch::TensorOptions().dtype(torch::kInt32).device(dev));
+    }
+    return mirror_index;
   }
 };
 
@@ -168,6 +191,86 @@ class ForwardSingleGPUfuseOp : public ntsGraphOp {
   }
 };
 
+/* Distributed fused aggregation op (ForwardGPUfuseOp surface,
+ * core/ntsDistGPUFusedGraphOp.hpp:48-91) — THE named hot-path operator, so
+ * toolkits/GCN.hpp:227 compiles unchanged against this layer.
+ *
+ * forward  ≙ Graph::sync_compute_decoupled (core/graph.hpp:3640-3719):
+ *   the P-step master→mirror ring.  Each rank owns the contiguous vertex
+ *   range partition_offset[r]..[r+1]; step s sends the rank's OWNED dense
+ *   feature block to rank (r−s)%P and receives rank (r+s)%P's block, then
+ *   aggregates that rank's chunk into the owned output.  Unlike the
+ *   reference there is no f_input.cpu() bounce (:58), no [vid|f×f32]
+ *   record packing (comm/network.cpp:476-495), no pinned-host spin queues:
+ *   dense fp32 blocks move GPU→GPU by grouped ncclSend/ncclRecv over xGMI,
+ *   stream-ordered with the aggregation kernels.
+ * backward ≙ Graph::compute_sync_decoupled (core/graph.hpp:3456-3622):
+ *   per remote partition k, the local chunk's CSR produces the partial
+ *   gradient block for k's masters; ring-send it to its owner, who
+ *   dense-adds (the reference's aggregate_data_buffer_debug merge,
+ *   cuda/ntsCUDATransferKernel.cuh:49-68, becomes a dense +=).
+ *
+ * P==1 degenerates to the local chunk only (no comm calls), matching the
+ * reference's single-rank pass-through (network.cpp:461-463).  The Python
+ * driver (neutronstarlite_amd/ring.py) implements the same protocol over
+ * torch.distributed with the overlap/pipelining variants; this C++ op is
+ * the link-contract implementation over the nts_comm C-ABI. */
+class ForwardGPUfuseOp : public ntsGraphOp {
+ public:
+  std::vector<CSC_segment_pinned *> subgraphs;
+  ForwardGPUfuseOp(PartitionedGraph *pg, VertexSubset *active)
+      : ntsGraphOp(pg, active) {
+    subgraphs = pg->graph_chunks;
+  }
+  NtsVar forward(NtsVar &f_input) override {
+    auto *pg = partitioned_graph_;
+    const int P = pg->partitions(), r = pg->partition_id;
+    const int64_t f = f_input.size(1);
+    assert(f_input.size(0) == (int64_t)pg->owned_vertices());
+    assert(P == 1 || pg->comm);
+    NtsVar x = f_input.contiguous();
+    NtsVar y = torch::zeros({(int64_t)pg->owned_vertices(), f}, x.options());
+    detail::csc_forward(pg->stream, *subgraphs[r], x, y);
+    for (int step = 1; step < P; step++) {
+      const int to = (r - step + P) % P, frm = (r + step) % P;
+      NtsVar recv = torch::empty({(int64_t)pg->part_n(frm), f}, x.options());
+      nts_comm_group_begin();
+      nts_comm_send_f32(pg->comm, pg->stream, x.data_ptr<float>(),
+                        x.numel(), to);
+      nts_comm_recv_f32(pg->comm, pg->stream, recv.data_ptr<float>(),
+                        recv.numel(), frm);
+      nts_comm_group_end();
+      detail::csc_forward(pg->stream, *subgraphs[frm], recv, y);
+    }
+    return y;
+  }
+  NtsVar backward(NtsVar &output_grad) override {
+    auto *pg = partitioned_graph_;
+    const int P = pg->partitions(), r = pg->partition_id;
+    const int64_t f = output_grad.size(1);
+    assert(output_grad.size(0) == (int64_t)pg->owned_vertices());
+    assert(P == 1 || pg->comm);
+    NtsVar g = output_grad.contiguous();
+    NtsVar gx = torch::zeros({(int64_t)pg->owned_vertices(), f}, g.options());
+    detail::csr_backward(pg->stream, *subgraphs[r], g, gx);
+    for (int step = 1; step < P; step++) {
+      const int k = (r + step) % P;          /* owner we feed */
+      const int peer_src = (r - step + P) % P; /* partial arriving for us */
+      NtsVar partial = torch::zeros({(int64_t)pg->part_n(k), f}, g.options());
+      detail::csr_backward(pg->stream, *subgraphs[k], g, partial);
+      NtsVar recv = torch::empty_like(gx);
+      nts_comm_group_begin();
+      nts_comm_send_f32(pg->comm, pg->stream, partial.data_ptr<float>(),
+                        partial.numel(), k);
+      nts_comm_recv_f32(pg->comm, pg->stream, recv.data_ptr<float>(),
+                        recv.numel(), peer_src);
+      nts_comm_group_end();
+      gx += recv;
+    }
+    return gx;
+  }
+};
+
 /* ---- decomposed edge-valued ops (GAT path, config #5) ----
  * The class names and signatures of core/ntsDistGPUGraphOp.hpp:48-361, so
  * GAT_GPU_DIST.hpp's 5-op chain (:191-215) compiles unchanged.  This host
@@ -179,13 +282,7 @@ class ForwardSingleGPUfuseOp : public ntsGraphOp {
 namespace detail {
 inline torch::Tensor &identity_mirror_index(PartitionedGraph *pg,
                                             torch::Device dev) {
-  static torch::Tensor idx;  /* single-GPU demo scope: whole-graph identity */
-  const int64_t v = pg->partition_offset.back();
-  if (!idx.defined() || idx.size(0) != v || idx.device() != dev) {
-    idx = torch::arange(v, torch::TensorOptions()
-                               .dtype(torch::kInt32).device(dev));
-  }
-  return idx;
+  return pg->mirror_index_for(dev);  /* per-graph state, not function-static */
 }
 }  // namespace detail
 
@@ -366,6 +463,18 @@ class NtsContext {
   NtsVar runVertexForward(std::function<NtsVar(NtsVar &)> f, NtsVar &input) {
     NtsVar in_leaf = input.detach().requires_grad_(true);
     NtsVar out = f(in_leaf);
+    tape_.push_back({NNOP, in_leaf, out, nullptr});
+    return out;
+  }
+
+  /* Two-input form used by toolkits/GCN.hpp:228-232: only nbr_input is
+   * taped (the reference's appendNNOp takes nbr_input,
+   * ntsContext.hpp:198-206); vtx_input contributes through torch autograd
+   * reachability only. */
+  NtsVar runVertexForward(std::function<NtsVar(NtsVar &, NtsVar &)> f,
+                          NtsVar &nbr_input, NtsVar &vtx_input) {
+    NtsVar in_leaf = nbr_input.detach().requires_grad_(true);
+    NtsVar out = f(in_leaf, vtx_input);
     tape_.push_back({NNOP, in_leaf, out, nullptr});
     return out;
   }
