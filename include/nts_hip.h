@@ -204,6 +204,12 @@ void nts_sample_reservoir(nts_stream *s, const nts_vid *column_offset,
     const nts_vid *row_indices, const nts_vid *dst_list, nts_vid n_dst,
     nts_vid fanout, unsigned long long seed, nts_vid *out_src,
     nts_vid *out_cnt);
+/* TEST-ONLY twin: forces the deterministic 64-bit (key,slot) tie fallback
+ * so GPU tests can exercise it; results must equal nts_sample_reservoir. */
+void nts_sample_reservoir_dbg_fallback(nts_stream *s,
+    const nts_vid *column_offset, const nts_vid *row_indices,
+    const nts_vid *dst_list, nts_vid n_dst, nts_vid fanout,
+    unsigned long long seed, nts_vid *out_src, nts_vid *out_cnt);
 
 /* gather-permute (additive): out[i] = in[index[i]] for f32 values and u32
  * indices — carries per-edge values between CSC and CSR edge order (e.g.

@@ -90,6 +90,7 @@ def lib():
     l.nts_permute_f32.argtypes = [_vp, _vp, _vp, _vp, _i64]
     l.nts_sample_reservoir.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32,
                                        _c.c_ulonglong, _vp, _vp]
+    l.nts_sample_reservoir_dbg_fallback.argtypes = l.nts_sample_reservoir.argtypes
     l.nts_device_count.restype = _i32
     l.nts_set_device.argtypes = [_i32]
     l.nts_build_arch.restype = _c.c_char_p
@@ -249,6 +250,14 @@ class Stream:
                                        _vp(row_indices), _vp(dst_list),
                                        n_dst, fanout, seed, _vp(out_src),
                                        _vp(out_cnt))
+
+    def sample_reservoir_dbg_fallback(self, column_offset, row_indices,
+                                      dst_list, n_dst, fanout, seed,
+                                      out_src, out_cnt):
+        """TEST-ONLY: forces the deterministic tie fallback."""
+        self._lib.nts_sample_reservoir_dbg_fallback(
+            self.h, _vp(column_offset), _vp(row_indices), _vp(dst_list),
+            n_dst, fanout, seed, _vp(out_src), _vp(out_cnt))
 
     def items_cache_clear(self):
         self._lib.nts_items_cache_clear(self.h)

@@ -78,6 +78,70 @@ def test_contract_and_determinism(setup):
         torch.equal(ly.row_indices_local, ly3.row_indices_local)
 
 
+def _hash_u32_host(seed, j):
+    """Host replica of k_hash_u32 (nts_hip.hip) for exactness checks."""
+    z = np.uint64(seed) ^ np.uint64(j)
+    with np.errstate(over="ignore"):
+        z = (z ^ (z >> np.uint64(33))) * np.uint64(0xFF51AFD7ED558CCD)
+        z = (z ^ (z >> np.uint64(33))) * np.uint64(0xC4CEB9FE1A85EC53)
+    return ((z ^ (z >> np.uint64(33))) & np.uint64(0xFFFFFFFF)).astype(np.uint32)
+
+
+def test_sampler_tie_fallback_deterministic_and_exact(setup):
+    """VERDICT/ADVICE r01: the tie fallback must be deterministic and select
+    the fanout smallest (key, slot) — not an atomic-race-ordered, capped
+    subset.  Exercised via the TEST-ONLY forced-fallback entry point on hub
+    destinations (deg up to 50k ≫ CAP=1024), checked against (a) the normal
+    path's output, (b) a host recomputation of the exact expected slots, and
+    (c) repeated-run equality."""
+    from neutronstarlite_amd import shim
+    dev = torch.device("cuda:0")
+    # adversarial graph: three hub destinations with huge in-degree
+    v = 60000
+    degs = [50000, 2000, 1500]
+    src_list, dst_list = [], []
+    rng = np.random.default_rng(17)
+    for d, deg in enumerate(degs):
+        src_list.append(rng.integers(0, v, size=deg, dtype=np.uint32))
+        dst_list.append(np.full(deg, d, dtype=np.uint32))
+    edges = np.stack([np.concatenate(src_list), np.concatenate(dst_list)],
+                     axis=1).astype(np.uint32)
+    outd, ind = G.degrees(edges, v)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    ch = G.build_chunks(edges, w, np.array([0, v], dtype=np.uint32), 0)[0]
+    from neutronstarlite_amd.ops import _u32_cuda
+    d_coff = _u32_cuda(ch.column_offset, dev)
+    d_rows = _u32_cuda(ch.row_indices, dev)
+    targets = torch.arange(3, dtype=torch.int32, device=dev)
+    n_dst, fanout, seed = 3, 13, 99
+    st = setup["st"]
+
+    def run(entry):
+        out_src = torch.zeros(n_dst * fanout, dtype=torch.int32, device=dev)
+        out_cnt = torch.zeros(n_dst, dtype=torch.int32, device=dev)
+        entry(d_coff.data_ptr(), d_rows.data_ptr(), targets.data_ptr(),
+              n_dst, fanout, seed, out_src.data_ptr(), out_cnt.data_ptr())
+        torch.cuda.synchronize()
+        return out_src.cpu().numpy(), out_cnt.cpu().numpy()
+
+    normal, cnt_n = run(st.sample_reservoir)
+    fb1, cnt_f = run(st.sample_reservoir_dbg_fallback)
+    fb2, _ = run(st.sample_reservoir_dbg_fallback)
+    assert np.array_equal(cnt_n, np.full(3, fanout))
+    assert np.array_equal(cnt_f, np.full(3, fanout))
+    assert np.array_equal(fb1, fb2), "fallback nondeterministic"
+    assert np.array_equal(normal, fb1), "fallback != threshold path"
+    # host-exact: the fanout smallest (key, slot-in-column) per destination
+    co = ch.column_offset
+    for i in range(n_dst):
+        e0, e1 = int(co[i]), int(co[i + 1])
+        keys = _hash_u32_host(seed, np.arange(e0, e1, dtype=np.uint64))
+        order = np.lexsort((np.arange(e1 - e0), keys))[:fanout]
+        expect = ch.row_indices[e0 + order]
+        got = normal[i * fanout:(i + 1) * fanout].astype(np.uint32)
+        assert np.array_equal(got, expect), f"dst {i}: not the k smallest keys"
+
+
 def test_gpu_sampled_aggregation_matches_oracle(setup):
     from neutronstarlite_amd.ops import HipEngine, MiniBatchFuseOp
     from neutronstarlite_amd.sampler_gpu import sample_subgraph_gpu
