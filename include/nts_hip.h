@@ -167,6 +167,36 @@ void nts_gather_msg_to_dst(nts_stream *s, float *dst_feature,
 void nts_edge_softmax_forward(nts_stream *s, float *msg_output,
     const float *msg_input, float *msg_cached, const nts_vid *row_indices,
     const nts_vid *column_offset, nts_vid batch_size, nts_vid feature_size);
+/* ---- additive GAT fusion entries (no reference twin; they fuse the
+ * permute / activation-mask / edge-dot passes the decomposed reference
+ * chain runs separately — semantics identical, verified in tests) ---- */
+
+/* Edge softmax ALSO emitted at perm_pos[e] (e.g. the CSC->CSR slot map),
+ * replacing a separate nts_permute_f32 pass. */
+void nts_edge_softmax_forward_dual(nts_stream *s, float *msg_output,
+    float *msg_output_perm, const nts_vid *perm_pos, const float *msg_input,
+    float *msg_cached, const nts_vid *column_offset, nts_vid batch_size,
+    nts_vid feature_size);
+
+/* Softmax backward with the leaky-relu derivative (lrelu_input[m] > 0 ? 1
+ * : slope) fused in, dual-emitted like the forward. */
+void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
+    float *msg_input_grad_perm, const nts_vid *perm_pos,
+    const float *msg_output_grad, const float *msg_cached,
+    const float *lrelu_input, float slope, const nts_vid *column_offset,
+    nts_vid batch_size, nts_vid feature_size);
+
+/* CSR backward gather that ALSO emits the per-edge dot
+ * dot_out[dot_pos[e]] = dot(input[column_indices[e]-dst_start], dot_vec[src])
+ * from the same streamed row bytes (the GAT attention-scalar gradient).
+ * Returns 1 if the fused kernel ran; 0 if the width forced the plain
+ * gather (caller must then run nts_edge_dot).  dot_pos NULL = identity. */
+int nts_gather_by_src_from_dst_dot(nts_stream *s, const float *input,
+    float *output, const float *weight_backward, const nts_vid *row_offset,
+    const nts_vid *column_indices, nts_vid dst_start, nts_vid batch_size,
+    nts_vid edges, nts_vid feature_size, const float *dot_vec, float *dot_out,
+    const nts_vid *dot_pos);
+
 void nts_edge_softmax_backward(nts_stream *s, float *msg_input_grad,
     const float *msg_output_grad, const float *msg_cached,
     const nts_vid *row_indices, const nts_vid *column_offset,

@@ -310,6 +310,108 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
   }
 }
 
+/* Fused CSR gather + per-edge dot (GAT backward): while the
+ * backward gather streams grad_y[dst(e)] row slabs to accumulate
+ * grad_h[src] = sum_e s[e]*grad_y[dst(e)], the SAME row bytes also feed
+ * gs[e] = dot(grad_y[dst(e)], h[src]) — the attention-scalar gradient the
+ * separate k_edge_dot kernel re-reads ~E*f floats to compute (9 ms of the
+ * 34 ms round-1 GAT step at Reddit scale).  h[src] is loaded once per
+ * (item, lane) into registers; the per-edge dot reduces across the lane
+ * group with xor shuffles and lane 0 writes gs to dot_pos[e] (the CSR->CSC
+ * slot map, so the softmax backward consumes it without a permute pass).
+ * Clean-case kernel: f % ELEM == 0 and f <= G*ELEM (one slab) — the
+ * launcher falls back to the unfused pair otherwise. */
+template <int ELEM>
+__global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm_dot(
+    const uint4 *__restrict__ items, const uint32_t *__restrict__ n_items_p,
+    const uint32_t *__restrict__ nbr, const float *__restrict__ ew,
+    const float *__restrict__ in, float *__restrict__ out, uint32_t nbr_start,
+    const float *__restrict__ dot_vec, float *__restrict__ dot_out,
+    const uint32_t *__restrict__ dot_pos, uint32_t f, uint32_t G) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t glane = threadIdx.x & (G - 1);
+  const uint32_t n_groups = (gridDim.x * blockDim.x) / G;
+  for (uint32_t it = tid / G; it < n_items; it += n_groups) {
+    const uint4 itm = items[it];
+    const uint32_t v = itm.x & 0x7fffffffu;
+    const bool shared_v = (itm.x >> 31) != 0;
+    const uint32_t e0 = itm.y, cnt = itm.z;
+    const uint32_t off = glane * ELEM;
+    const bool act = off + ELEM <= f; /* f%ELEM==0: lane is full or idle */
+    float acc[ELEM] = {};
+    float hreg[ELEM] = {};
+    if (act) {
+      const float *hp = dot_vec + (uint64_t)v * f + off;
+#pragma unroll
+      for (int j = 0; j < ELEM; ++j) hreg[j] = hp[j];
+    }
+    uint32_t e = e0;
+    const uint32_t e_end = e0 + cnt;
+    for (; e + 4 <= e_end; e += 4) {
+      float dp[4] = {};
+      if (act) {
+        const float *p[4];
+        float wv[4];
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          p[k] = in + (uint64_t)(nbr[e + k] - nbr_start) * f + off;
+          wv[k] = ew[e + k];
+        }
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+#pragma unroll
+          for (int j = 0; j < ELEM; ++j) {
+            const float x = p[k][j];
+            acc[j] = fmaf(wv[k], x, acc[j]);
+            dp[k] = fmaf(hreg[j], x, dp[k]);
+          }
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+#pragma unroll
+        for (int w = 32; w >= 1; w >>= 1) {
+          if ((uint32_t)w < G) dp[k] += __shfl_xor(dp[k], w, 64);
+        }
+      }
+      if (glane == 0) {
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+          dot_out[dot_pos ? dot_pos[e + k] : e + k] = dp[k];
+      }
+    }
+    for (; e < e_end; ++e) {
+      float dp = 0.f;
+      if (act) {
+        const float w = ew[e];
+        const float *p = in + (uint64_t)(nbr[e] - nbr_start) * f + off;
+#pragma unroll
+        for (int j = 0; j < ELEM; ++j) {
+          const float x = p[j];
+          acc[j] = fmaf(w, x, acc[j]);
+          dp = fmaf(hreg[j], x, dp);
+        }
+      }
+#pragma unroll
+      for (int w = 32; w >= 1; w >>= 1) {
+        if ((uint32_t)w < G) dp += __shfl_xor(dp, w, 64);
+      }
+      if (glane == 0) dot_out[dot_pos ? dot_pos[e] : e] = dp;
+    }
+    if (act) {
+      float *o = out + (uint64_t)v * f + off;
+      if (!shared_v) {
+#pragma unroll
+        for (int j = 0; j < ELEM; ++j) o[j] += acc[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < ELEM; ++j) atomicAdd(&o[j], acc[j]);
+      }
+    }
+  }
+}
+
 /* ---- GPU-resident fan-out sampling ----
  * One wave per destination.  Contract of Sampler::reservoir_sample
  * (ntsSampler.hpp:113-166): min(deg, fanout) uniformly chosen in-edge
@@ -580,13 +682,24 @@ __global__ void k_edge_softmax_sum(const uint4 *__restrict__ items,
   }
 }
 
+/* Normalize pass.  Optional extras (GAT fusions):
+ *  - out2/pos2: ALSO write each value to out2[pos2[m]] — with pos2 = the
+ *    CSC->CSR slot map this emits the result in CSR edge order in the same
+ *    pass, replacing a separate nts_permute_f32 kernel (2x ~2 ms/step on
+ *    the Reddit-scale GAT config).
+ *  - lrelu_in/slope (BACKWARD only): multiply by the leaky-relu derivative
+ *    (lrelu_in[m] > 0 ? 1 : slope), fusing the attention activation's
+ *    backward elementwise pass. */
 template <bool BACKWARD>
 __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
                                     const uint32_t *__restrict__ n_items_p,
                                     float *__restrict__ out,
                                     const float *__restrict__ cached,
                                     const float *__restrict__ sums,
-                                    uint32_t f) {
+                                    float *__restrict__ out2,
+                                    const uint32_t *__restrict__ pos2,
+                                    const float *__restrict__ lrelu_in,
+                                    float slope, uint32_t f) {
   const uint32_t n_items = *n_items_p;
   const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const uint32_t lane = threadIdx.x & 63;
@@ -601,8 +714,15 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
       const uint32_t r = (uint32_t)(i % f);
       const uint64_t m = (uint64_t)e * f + r;
       const float sum = sums[(uint64_t)d * f + r];
-      if (BACKWARD) out[m] = out[m] - sum * cached[m];
-      else out[m] = out[m] / sum;
+      float v;
+      if (BACKWARD) {
+        v = out[m] - sum * cached[m];
+        if (lrelu_in) v *= (lrelu_in[m] > 0.f) ? 1.f : slope;
+      } else {
+        v = out[m] / sum;
+      }
+      out[m] = v;
+      if (out2) out2[(uint64_t)pos2[e] * f + r] = v; /* pos2: per-edge map */
     }
   }
 }
@@ -1063,7 +1183,10 @@ static void launch_edge_op(nts_stream *s, EdgeOp op, float *message,
 static void launch_edge_softmax(nts_stream *s, bool backward, float *out,
                                 const float *in, const float *cached,
                                 const nts_vid *column_offset, nts_vid batch,
-                                nts_vid f) {
+                                nts_vid f, float *out2 = nullptr,
+                                const nts_vid *pos2 = nullptr,
+                                const float *lrelu_in = nullptr,
+                                float slope = 0.f) {
   if (!batch || !f) return;
   const uint32_t edges = read_edge_count(s, column_offset, batch);
   if (!edges) return;
@@ -1077,14 +1200,14 @@ static void launch_edge_softmax(nts_stream *s, bool backward, float *out,
                        sums, f);
     hipLaunchKernelGGL((k_edge_softmax_norm<true>), dim3(grid),
                        dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
-                       out, cached, sums, f);
+                       out, cached, sums, out2, pos2, lrelu_in, slope, f);
   } else {
     hipLaunchKernelGGL((k_edge_softmax_sum<false>), dim3(grid),
                        dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
                        out, in, cached, sums, f);
     hipLaunchKernelGGL((k_edge_softmax_norm<false>), dim3(grid),
                        dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
-                       out, cached, sums, f);
+                       out, cached, sums, out2, pos2, nullptr, 0.f, f);
   }
   dbg_sync(s, "k_edge_softmax");
 }
@@ -1208,6 +1331,86 @@ void nts_edge_softmax_backward(nts_stream *s, float *msg_input_grad,
   (void)row_indices;
   launch_edge_softmax(s, true, msg_input_grad, msg_output_grad, msg_cached,
                       column_offset, batch_size, feature_size);
+}
+
+/* ---- GAT fusion entry points (additive; see include/nts_hip.h) ---- */
+
+void nts_edge_softmax_forward_dual(nts_stream *s, float *msg_output,
+                                   float *msg_output_perm,
+                                   const nts_vid *perm_pos,
+                                   const float *msg_input, float *msg_cached,
+                                   const nts_vid *column_offset,
+                                   nts_vid batch_size, nts_vid feature_size) {
+  launch_edge_softmax(s, false, msg_output, msg_input, nullptr, column_offset,
+                      batch_size, feature_size, msg_output_perm, perm_pos);
+  if (msg_cached && msg_cached != msg_output && batch_size && feature_size) {
+    const uint32_t edges = read_edge_count(s, column_offset, batch_size);
+    const uint64_t n = (uint64_t)edges * feature_size;
+    if (n)
+      hipLaunchKernelGGL(k_copy, dim3(grid_for(n)), dim3(NTS_BLOCK), 0,
+                         s->stream, msg_cached, msg_output, n);
+  }
+}
+
+void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
+                                     float *msg_input_grad_perm,
+                                     const nts_vid *perm_pos,
+                                     const float *msg_output_grad,
+                                     const float *msg_cached,
+                                     const float *lrelu_input, float slope,
+                                     const nts_vid *column_offset,
+                                     nts_vid batch_size, nts_vid feature_size) {
+  launch_edge_softmax(s, true, msg_input_grad, msg_output_grad, msg_cached,
+                      column_offset, batch_size, feature_size,
+                      msg_input_grad_perm, perm_pos, lrelu_input, slope);
+}
+
+int nts_gather_by_src_from_dst_dot(nts_stream *s, const float *input,
+                                   float *output, const float *weight_backward,
+                                   const nts_vid *row_offset,
+                                   const nts_vid *column_indices,
+                                   nts_vid dst_start, nts_vid batch_size,
+                                   nts_vid edges, nts_vid feature_size,
+                                   const float *dot_vec, float *dot_out,
+                                   const nts_vid *dot_pos) {
+  const uint32_t f = feature_size;
+  if (!batch_size || !edges || !f) return 1;
+  const uintptr_t a =
+      (uintptr_t)input | (uintptr_t)output | (uintptr_t)dot_vec;
+  int elem;
+  if (f % 4 == 0 && a % 16 == 0) elem = 4;
+  else if (f % 2 == 0 && a % 8 == 0) elem = 2;
+  else elem = 1;
+  uint32_t G = 64;
+  if (elem > 1) {
+    const uint32_t need = (f + elem - 1) / elem;
+    while (G / 2 >= need && G > 16) G /= 2;
+  }
+  if (elem == 1 || f > G * (uint32_t)elem) {
+    /* ragged or multi-slab width: unfused fallback — plain CSR gather; the
+     * caller must compute the dot with nts_edge_dot */
+    launch_gather(s, input, output, weight_backward, column_indices,
+                  row_offset, dst_start, batch_size, edges, f, 1,
+                  NTS_KTAG_BWD);
+    return 0;
+  }
+  ItemsBuf &ib = get_items(s, row_offset, batch_size, edges);
+  const uint64_t bound_groups = (uint64_t)batch_size + edges / NTS_SPLIT + 1;
+  const uint32_t grid = grid_for(bound_groups * G);
+  Tic t(s, NTS_KTAG_BWD);
+  if (elem == 4) {
+    hipLaunchKernelGGL((k_gather_spmm_dot<4>), dim3(grid), dim3(NTS_BLOCK), 0,
+                       s->stream, ib.items, ib.counter, column_indices,
+                       weight_backward, input, output, dst_start, dot_vec,
+                       dot_out, dot_pos, f, G);
+  } else {
+    hipLaunchKernelGGL((k_gather_spmm_dot<2>), dim3(grid), dim3(NTS_BLOCK), 0,
+                       s->stream, ib.items, ib.counter, column_indices,
+                       weight_backward, input, output, dst_start, dot_vec,
+                       dot_out, dot_pos, f, G);
+  }
+  dbg_sync(s, "k_gather_spmm_dot");
+  return 1;
 }
 
 void nts_sample_reservoir(nts_stream *s, const nts_vid *column_offset,

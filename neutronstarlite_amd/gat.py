@@ -48,6 +48,12 @@ class GATLayer:
         self.csr_from_csc = torch.from_numpy(perm).to(device)
         self._perm_u32 = torch.from_numpy(
             perm.astype(np.uint32).view(np.int32)).to(device)
+        # inverse map (CSC slot -> CSR slot) for the dual-order softmax
+        # emission (kernels write out2[pos[e]] while walking CSC items)
+        inv = np.empty_like(perm)
+        inv[perm] = np.arange(len(perm))
+        self._inv_perm_u32 = torch.from_numpy(
+            inv.astype(np.uint32).view(np.int32)).to(device)
         self._doe = self._soe = None
         self._ones_dst = self._ones_src = None
         # separate wrapper of the same HIP stream for the scalar (f=1)
@@ -86,17 +92,24 @@ class GATLayer:
         m_sum = m_src + m_dst
         e = torch.nn.functional.leaky_relu(m_sum, negative_slope)
         s = torch.empty(E, 1, device=dev)
+        s_csr = torch.empty(E, 1, device=dev)
         cached = torch.empty(E, 1, device=dev)
-        st.edge_softmax_forward(s.data_ptr(), e.contiguous().data_ptr(),
-                                cached.data_ptr(), ch.row_indices.data_ptr(),
-                                ch.column_offset.data_ptr(), ch.dst_n, 1)
+        # dual emission: softmax output lands in CSC order (s) AND directly
+        # in CSR edge order (s_csr) in the same normalize pass, replacing
+        # the backward's nts_permute_f32 (round-1: 2.1 ms/step)
+        st.edge_softmax_forward_dual(s.data_ptr(), s_csr.data_ptr(),
+                                     self._inv_perm_u32.data_ptr(),
+                                     e.contiguous().data_ptr(),
+                                     cached.data_ptr(),
+                                     ch.column_offset.data_ptr(), ch.dst_n, 1)
         y = torch.zeros(ch.dst_n, h.shape[1], device=dev)
         st.gather_by_dst_from_src(h.data_ptr(), y.data_ptr(), s.data_ptr(),
                                   ch.row_indices.data_ptr(),
                                   ch.column_offset.data_ptr(),
                                   ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
                                   E, ch.dst_n, h.shape[1], with_weight=True)
-        saved = {"h": h, "s": s, "cached": cached, "m_sum": m_sum}
+        saved = {"h": h, "s": s, "s_csr": s_csr, "cached": cached,
+                 "m_sum": m_sum}
         return y, saved
 
     def backward(self, grad_y: torch.Tensor, saved,
@@ -106,31 +119,34 @@ class GATLayer:
         dev = grad_y.device
         f = grad_y.shape[1]
         grad_y = grad_y.contiguous()
-        # CSR gather with attention weights permuted into CSR edge order
-        # (nts_permute_f32: ~8x torch's int64 index_select at this size)
-        s_bwd = torch.empty(E, 1, device=dev)
-        st.permute_f32(s_bwd.data_ptr(), saved["s"].data_ptr(),
-                       self._perm_u32.data_ptr(), E)
-        grad_h = torch.zeros(ch.src_n, f, device=dev)
-        st.gather_by_src_from_dst(grad_y.data_ptr(), grad_h.data_ptr(),
-                                  s_bwd.data_ptr(),
-                                  ch.row_offset.data_ptr(),
-                                  ch.column_indices.data_ptr(),
-                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
-                                  E, ch.src_n, f, with_weight=True)
-        # d y / d s[e] = grad_y[dst(e)] . h[src(e)] — fused edge-dot kernel
-        # (E scalars; never materializes E x f edge tensors)
+        # fused CSR gather + edge-dot: grad_h[src] accumulates while the
+        # SAME grad_y row bytes produce gs[e] = grad_y[dst(e)].h[src(e)],
+        # written straight into CSC slot order (dot_pos = csr->csc map) for
+        # the softmax backward.  Replaces the separate k_edge_dot pass
+        # (round-1: ~9 ms/step at f=128) and the s permute (dual-order
+        # softmax output from forward).  rc=0 -> width not fused (ragged /
+        # multi-slab f): the plain gather ran; do the dot separately.
         gs = torch.empty(E, 1, device=dev)
-        st.edge_dot(gs.data_ptr(), grad_y.data_ptr(),
-                    saved["h"].data_ptr(), ch.row_indices.data_ptr(),
-                    ch.column_offset.data_ptr(), ch.src_s, ch.dst_n, f)
+        grad_h = torch.zeros(ch.src_n, f, device=dev)
+        rc = st.gather_by_src_from_dst_dot(
+            grad_y.data_ptr(), grad_h.data_ptr(), saved["s_csr"].data_ptr(),
+            ch.row_offset.data_ptr(), ch.column_indices.data_ptr(),
+            ch.dst_s, ch.src_n, E, f, saved["h"].data_ptr(), gs.data_ptr(),
+            self._perm_u32.data_ptr())
+        if rc == 0:
+            st.edge_dot(gs.data_ptr(), grad_y.data_ptr(),
+                        saved["h"].data_ptr(), ch.row_indices.data_ptr(),
+                        ch.column_offset.data_ptr(), ch.src_s, ch.dst_n, f)
+        # softmax backward with the leaky-relu derivative fused in and the
+        # result dual-emitted in CSC (ge) and CSR (ge_csr) edge order —
+        # replaces the second permute and the torch elementwise mask pass
         ge = torch.empty(E, 1, device=dev)
-        st.edge_softmax_backward(ge.data_ptr(), gs.contiguous().data_ptr(),
-                                 saved["cached"].data_ptr(),
-                                 ch.row_indices.data_ptr(),
-                                 ch.column_offset.data_ptr(), ch.dst_n, 1)
-        ge = ge * torch.where(saved["m_sum"] > 0, 1.0, negative_slope)
-        ge = ge.contiguous()
+        ge_csr = torch.empty(E, 1, device=dev)
+        st.edge_softmax_backward_fused(
+            ge.data_ptr(), ge_csr.data_ptr(), self._inv_perm_u32.data_ptr(),
+            gs.data_ptr(), saved["cached"].data_ptr(),
+            saved["m_sum"].contiguous().data_ptr(), negative_slope,
+            ch.column_offset.data_ptr(), ch.dst_n, 1)
         # edge-scalar -> vertex reductions through the load-balanced gather
         # kernel (per-edge values as WEIGHTS over an all-ones input): the
         # direct msg->vertex atomics serialize on power-law hub sources
@@ -140,9 +156,6 @@ class GATLayer:
         if self._ones_dst is None:
             self._ones_dst = torch.ones(ch.dst_n, 1, device=dev)
             self._ones_src = torch.ones(ch.src_n, 1, device=dev)
-        ge_csr = torch.empty(E, 1, device=dev)
-        st.permute_f32(ge_csr.data_ptr(), ge.data_ptr(),
-                       self._perm_u32.data_ptr(), E)
         sst = self.scalar_stream
         g_src = torch.zeros(ch.src_n, 1, device=dev)
         sst.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),

@@ -87,6 +87,12 @@ def lib():
     l.nts_edge_dot.argtypes = [_vp] + [_vp] * 5 + [_u32] * 3
     l.nts_edge_softmax_forward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
+    l.nts_edge_softmax_forward_dual.argtypes = [_vp] + [_vp] * 6 + [_u32] * 2
+    l.nts_edge_softmax_backward_fused.argtypes = (
+        [_vp] + [_vp] * 6 + [_c.c_float] + [_vp] + [_u32] * 2)
+    l.nts_gather_by_src_from_dst_dot.argtypes = (
+        [_vp] + [_vp] * 5 + [_u32] * 4 + [_vp] * 3)
+    l.nts_gather_by_src_from_dst_dot.restype = _i32
     l.nts_permute_f32.argtypes = [_vp, _vp, _vp, _vp, _i64]
     l.nts_sample_reservoir.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32,
                                        _c.c_ulonglong, _vp, _vp]
@@ -243,6 +249,28 @@ class Stream:
 
     def permute_f32(self, out, inp, index, n):
         self._lib.nts_permute_f32(self.h, _vp(out), _vp(inp), _vp(index), n)
+
+    def edge_softmax_forward_dual(self, out, out_perm, perm_pos, inp, cached,
+                                  column_offset, batch, f):
+        self._lib.nts_edge_softmax_forward_dual(
+            self.h, _vp(out), _vp(out_perm), _vp(perm_pos), _vp(inp),
+            _vp(cached), _vp(column_offset), batch, f)
+
+    def edge_softmax_backward_fused(self, in_grad, in_grad_perm, perm_pos,
+                                    out_grad, cached, lrelu_input, slope,
+                                    column_offset, batch, f):
+        self._lib.nts_edge_softmax_backward_fused(
+            self.h, _vp(in_grad), _vp(in_grad_perm), _vp(perm_pos),
+            _vp(out_grad), _vp(cached), _vp(lrelu_input), slope,
+            _vp(column_offset), batch, f)
+
+    def gather_by_src_from_dst_dot(self, inp, out, weight, row_offset,
+                                   column_indices, dst_start, batch, edges, f,
+                                   dot_vec, dot_out, dot_pos):
+        return self._lib.nts_gather_by_src_from_dst_dot(
+            self.h, _vp(inp), _vp(out), _vp(weight), _vp(row_offset),
+            _vp(column_indices), dst_start, batch, edges, f, _vp(dot_vec),
+            _vp(dot_out), _vp(dot_pos))
 
     def sample_reservoir(self, column_offset, row_indices, dst_list, n_dst,
                          fanout, seed, out_src, out_cnt):

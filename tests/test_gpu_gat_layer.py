@@ -36,10 +36,13 @@ def torch_gat_reference(h, a_src, a_dst, dst_of_edge, src_of_edge, v, slope):
     return y
 
 
-def test_gat_layer_forward_backward():
+@pytest.mark.parametrize("f", [32, 33, 128])
+def test_gat_layer_forward_backward(f):
+    """f=32/128 run the fused CSR-gather+dot path; f=33 (odd) exercises the
+    unfused fallback (rc=0 -> separate edge-dot) through the same layer."""
     from neutronstarlite_amd.gat import GATLayer
     dev = torch.device("cuda:0")
-    v, e, f, slope = 1200, 20000, 32, 0.2
+    v, e, slope = 1200, 20000, 0.2
     edges = G.rmat_edges(v, e, seed=13)
     outd, ind = G.degrees(edges, v)
     w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
