@@ -90,6 +90,16 @@ def partition_offsets(edges: np.ndarray, v: int, parts: int,
         target = total * p / parts
         offs.append(int(np.searchsorted(csum, target)))
     offs.append(v)
+    # Hub-dominated graphs can make searchsorted return duplicate
+    # boundaries; degrade to a valid (unbalanced) partition by nudging each
+    # boundary past its predecessor instead of asserting.
+    # Requires v >= parts (checked) so every range stays non-empty.
+    assert v >= parts, f"{parts} partitions need >= {parts} vertices"
+    for p in range(1, parts + 1):
+        lo = offs[p - 1] + 1
+        hi = v - (parts - p)
+        offs[p] = min(max(offs[p], lo), hi)
+    offs[parts] = v
     offs = np.array(offs, dtype=np.uint32)
     assert np.all(np.diff(offs.astype(np.int64)) > 0), "empty partition"
     return offs

@@ -97,11 +97,19 @@ def setup_mirror_lists(rg: "RingGraph"):
     for k in range(P):
         if k == r:
             continue
-        # we tell rank k what we need from it; rank k tells us what it needs
-        send_ids = need[k].to(dev)
-        recv_ids = torch.zeros(int(lens[k, r].item()), dtype=torch.int64,
-                               device=dev)
-        reqs += _exchange(send_ids, k, recv_ids, k)
+        # we tell rank k what we need from it; rank k tells us what it needs.
+        # Zero-element P2P ops are SKIPPED (empty NCCL send/recv is
+        # version-fragile); both sides see the all-reduced lens
+        # matrix, so the skips pair up.
+        ops = []
+        if len(need[k]):
+            ops.append(dist.P2POp(dist.isend, need[k].to(dev), k))
+        n_recv = int(lens[k, r].item())
+        recv_ids = torch.zeros(n_recv, dtype=torch.int64, device=dev)
+        if n_recv:
+            ops.append(dist.P2POp(dist.irecv, recv_ids, k))
+        if ops:
+            reqs += dist.batch_isend_irecv(ops)
         serve[k] = recv_ids
     for rq in reqs:
         rq.wait()
@@ -125,10 +133,17 @@ def _ring_forward_filtered(rg: "RingGraph", x_owned, engine):
         return x_owned.index_select(0, rows).contiguous()
 
     def post(step):
+        # zero-element P2P ops are skipped (both sides know the static list
+        # lengths, so the skips pair up)
         to = (r - step) % P
         frm = (r + step) % P
         recv = torch.empty(len(rg.need[frm]), f, device=dev)
-        reqs = _exchange(pack_for(to), to, recv, frm)
+        ops = []
+        if len(rg.serve[to]):
+            ops.append(dist.P2POp(dist.isend, pack_for(to), to))
+        if len(rg.need[frm]):
+            ops.append(dist.P2POp(dist.irecv, recv, frm))
+        reqs = dist.batch_isend_irecv(ops) if ops else []
         return reqs, recv, frm
 
     pending = post(1)
@@ -139,9 +154,10 @@ def _ring_forward_filtered(rg: "RingGraph", x_owned, engine):
             nxt = post(step + 1)
         for rq in reqs:
             rq.wait()
-        dense = torch.zeros(rg.part_n(frm), f, device=dev)
-        dense.index_copy_(0, rg.need[frm] - rg.offs[frm], recv)
-        engine.csc_forward(rg.chunks[frm], dense, y)
+        if len(rg.need[frm]):
+            dense = torch.zeros(rg.part_n(frm), f, device=dev)
+            dense.index_copy_(0, rg.need[frm] - rg.offs[frm], recv)
+            engine.csc_forward(rg.chunks[frm], dense, y)
         if step + 1 < P:
             pending = nxt
     return y

@@ -34,6 +34,21 @@ def test_partition_offsets_balance():
     assert max(loads) < 2.0 * (sum(loads) / 4)
 
 
+def test_partition_offsets_hub_dominated():
+    """One vertex carrying almost all load must yield a valid (if
+    unbalanced) partition, not an assertion: duplicate
+    searchsorted boundaries are nudged forward."""
+    v, parts = 64, 4
+    src = np.concatenate([np.zeros(10000, dtype=np.uint32),
+                          np.arange(v, dtype=np.uint32)])
+    dst = np.concatenate([np.arange(10000, dtype=np.uint32) % v,
+                          np.arange(v, dtype=np.uint32)])
+    edges = np.stack([src, dst], axis=1)
+    offs = G.partition_offsets(edges, v, parts)
+    d = np.diff(offs.astype(np.int64))
+    assert offs[0] == 0 and offs[-1] == v and (d > 0).all()
+
+
 def test_chunks_cover_all_edges():
     v, parts = 1024, 4
     edges = G.rmat_edges(v, 20000, seed=7)
