@@ -481,7 +481,11 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
     }
     unsigned long long T64 = 0;
     if (landed) {
-      T64 = T << 32; /* (key,slot) candidates below T<<32 == keys below T */
+      /* (key,slot) candidates below T<<32 == keys below T.  T can exceed
+       * 2^32 when deg < 2*fanout (the initial threshold over-covers and
+       * lands immediately): clamp to all-ones instead of overflowing the
+       * shift to zero (which collected nothing and read garbage slots). */
+      T64 = (T >= 0x100000000ULL) ? ~0ULL : (T << 32);
     } else {
       /* Massive key ties: the 32-bit survivor count jumps over the
        * [fanout, cap] window.  Search the 64-bit candidate (key<<32)|slot
