@@ -179,12 +179,15 @@ void nts_edge_softmax_forward_dual(nts_stream *s, float *msg_output,
     nts_vid feature_size);
 
 /* Softmax backward with the leaky-relu derivative (lrelu_input[m] > 0 ? 1
- * : slope) fused in, dual-emitted like the forward. */
+ * : slope) fused in, dual-emitted like the forward.  dst_sum (optional,
+ * f==1 only, caller-zeroed): ALSO accumulates the per-destination sum of
+ * the result — the attention scalar's dst gradient — saving a separate
+ * f=1 reduction pass. */
 void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
     float *msg_input_grad_perm, const nts_vid *perm_pos,
     const float *msg_output_grad, const float *msg_cached,
-    const float *lrelu_input, float slope, const nts_vid *column_offset,
-    nts_vid batch_size, nts_vid feature_size);
+    const float *lrelu_input, float slope, float *dst_sum,
+    const nts_vid *column_offset, nts_vid batch_size, nts_vid feature_size);
 
 /* CSR backward gather that ALSO emits the per-edge dot
  * dot_out[dot_pos[e]] = dot(input[column_indices[e]-dst_start], dot_vec[src])

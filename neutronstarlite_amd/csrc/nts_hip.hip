@@ -703,7 +703,9 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
                                     float *__restrict__ out2,
                                     const uint32_t *__restrict__ pos2,
                                     const float *__restrict__ lrelu_in,
-                                    float slope, uint32_t f) {
+                                    float slope,
+                                    float *__restrict__ dst_accum,
+                                    uint32_t f) {
   const uint32_t n_items = *n_items_p;
   const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const uint32_t lane = threadIdx.x & 63;
@@ -713,6 +715,7 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
     const uint32_t d = itm.x & 0x7fffffffu;
     const uint32_t e0 = itm.y;
     const uint64_t nel = (uint64_t)itm.z * f;
+    float gsum = 0.f; /* per-dst sum of the result (f==1 only) */
     for (uint64_t i = lane; i < nel; i += 64) {
       const uint32_t e = e0 + (uint32_t)(i / f);
       const uint32_t r = (uint32_t)(i % f);
@@ -727,6 +730,15 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
       }
       out[m] = v;
       if (out2) out2[(uint64_t)pos2[e] * f + r] = v; /* pos2: per-edge map */
+      if (dst_accum) gsum += v;
+    }
+    if (dst_accum) {
+      /* the attention-scalar's per-destination reduction (gat.py g_dst),
+       * emitted here instead of a separate f=1 gather pass; item-bounded,
+       * so one fp32 atomic per (item, wave), hub-safe */
+#pragma unroll
+      for (int w = 32; w >= 1; w >>= 1) gsum += __shfl_xor(gsum, w, 64);
+      if (lane == 0) atomicAdd(&dst_accum[d], gsum);
     }
   }
 }
@@ -1190,7 +1202,9 @@ static void launch_edge_softmax(nts_stream *s, bool backward, float *out,
                                 nts_vid f, float *out2 = nullptr,
                                 const nts_vid *pos2 = nullptr,
                                 const float *lrelu_in = nullptr,
-                                float slope = 0.f) {
+                                float slope = 0.f,
+                                float *dst_accum = nullptr) {
+  if (dst_accum && f != 1) abort(); /* per-dst sum emission is f==1 only */
   if (!batch || !f) return;
   const uint32_t edges = read_edge_count(s, column_offset, batch);
   if (!edges) return;
@@ -1204,14 +1218,16 @@ static void launch_edge_softmax(nts_stream *s, bool backward, float *out,
                        sums, f);
     hipLaunchKernelGGL((k_edge_softmax_norm<true>), dim3(grid),
                        dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
-                       out, cached, sums, out2, pos2, lrelu_in, slope, f);
+                       out, cached, sums, out2, pos2, lrelu_in, slope,
+                       dst_accum, f);
   } else {
     hipLaunchKernelGGL((k_edge_softmax_sum<false>), dim3(grid),
                        dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
                        out, in, cached, sums, f);
     hipLaunchKernelGGL((k_edge_softmax_norm<false>), dim3(grid),
                        dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
-                       out, cached, sums, out2, pos2, nullptr, 0.f, f);
+                       out, cached, sums, out2, pos2, nullptr, 0.f,
+                       dst_accum, f);
   }
   dbg_sync(s, "k_edge_softmax");
 }
@@ -1362,11 +1378,13 @@ void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
                                      const float *msg_output_grad,
                                      const float *msg_cached,
                                      const float *lrelu_input, float slope,
+                                     float *dst_sum,
                                      const nts_vid *column_offset,
                                      nts_vid batch_size, nts_vid feature_size) {
   launch_edge_softmax(s, true, msg_input_grad, msg_output_grad, msg_cached,
                       column_offset, batch_size, feature_size,
-                      msg_input_grad_perm, perm_pos, lrelu_input, slope);
+                      msg_input_grad_perm, perm_pos, lrelu_input, slope,
+                      dst_sum);
 }
 
 int nts_gather_by_src_from_dst_dot(nts_stream *s, const float *input,

@@ -142,20 +142,27 @@ class GATLayer:
         # replaces the second permute and the torch elementwise mask pass
         ge = torch.empty(E, 1, device=dev)
         ge_csr = torch.empty(E, 1, device=dev)
+        g_dst = torch.zeros(ch.dst_n, 1, device=dev)
         st.edge_softmax_backward_fused(
             ge.data_ptr(), ge_csr.data_ptr(), self._inv_perm_u32.data_ptr(),
             gs.data_ptr(), saved["cached"].data_ptr(),
             saved["m_sum"].contiguous().data_ptr(), negative_slope,
-            ch.column_offset.data_ptr(), ch.dst_n, 1)
+            ch.column_offset.data_ptr(), ch.dst_n, 1,
+            dst_sum=g_dst.data_ptr())
         # edge-scalar -> vertex reductions through the load-balanced gather
         # kernel (per-edge values as WEIGHTS over an all-ones input): the
         # direct msg->vertex atomics serialize on power-law hub sources
         # (one address takes every hub edge's atomicAdd — measured 25 ms of
         # a 115 ms step), while the gather's work items split hubs and merge
         # a handful of partials.
+        # g_dst came out of the softmax-backward pass itself (dst_sum);
+        # g_src still needs the src-major reduction — through the
+        # load-balanced gather kernel (per-edge values as WEIGHTS over an
+        # all-ones input): direct msg->vertex atomics serialize on
+        # power-law hub sources (one address takes every hub edge's
+        # atomicAdd — measured 25 ms of a 115 ms step in round 1).
         if self._ones_dst is None:
             self._ones_dst = torch.ones(ch.dst_n, 1, device=dev)
-            self._ones_src = torch.ones(ch.src_n, 1, device=dev)
         sst = self.scalar_stream
         g_src = torch.zeros(ch.src_n, 1, device=dev)
         sst.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),
@@ -163,10 +170,4 @@ class GATLayer:
                                   ch.column_indices.data_ptr(),
                                   ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
                                   E, ch.src_n, 1, with_weight=True)
-        g_dst = torch.zeros(ch.dst_n, 1, device=dev)
-        sst.gather_by_dst_from_src(self._ones_src.data_ptr(), g_dst.data_ptr(),
-                                  ge.data_ptr(), ch.row_indices.data_ptr(),
-                                  ch.column_offset.data_ptr(),
-                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
-                                  E, ch.dst_n, 1, with_weight=True)
         return grad_h, g_src[:, 0], g_dst[:, 0]

@@ -89,7 +89,7 @@ def lib():
     l.nts_edge_softmax_backward.argtypes = [_vp] + [_vp] * 5 + [_u32] * 2
     l.nts_edge_softmax_forward_dual.argtypes = [_vp] + [_vp] * 6 + [_u32] * 2
     l.nts_edge_softmax_backward_fused.argtypes = (
-        [_vp] + [_vp] * 6 + [_c.c_float] + [_vp] + [_u32] * 2)
+        [_vp] + [_vp] * 6 + [_c.c_float] + [_vp] * 2 + [_u32] * 2)
     l.nts_gather_by_src_from_dst_dot.argtypes = (
         [_vp] + [_vp] * 5 + [_u32] * 4 + [_vp] * 3)
     l.nts_gather_by_src_from_dst_dot.restype = _i32
@@ -258,11 +258,11 @@ class Stream:
 
     def edge_softmax_backward_fused(self, in_grad, in_grad_perm, perm_pos,
                                     out_grad, cached, lrelu_input, slope,
-                                    column_offset, batch, f):
+                                    column_offset, batch, f, dst_sum=None):
         self._lib.nts_edge_softmax_backward_fused(
             self.h, _vp(in_grad), _vp(in_grad_perm), _vp(perm_pos),
             _vp(out_grad), _vp(cached), _vp(lrelu_input), slope,
-            _vp(column_offset), batch, f)
+            _vp(dst_sum), _vp(column_offset), batch, f)
 
     def gather_by_src_from_dst_dot(self, inp, out, weight, row_offset,
                                    column_indices, dst_start, batch, edges, f,
