@@ -210,6 +210,46 @@ def dep_nbr_backward(dg: DepGraph, mirror_grad: torch.Tensor) -> torch.Tensor:
     return gx
 
 
+class DistGATLayerGPU:
+    """Product path for P>=1 on GPU: dep-neighbor mirror exchange composed
+    with the HIP GATLayer running on the REINDEXED (mirror-slot) chunk.
+    Because the reindexed CSC/CSR is just a graph whose "source vertices"
+    are the mirror slots, the single-GPU attention kernels (gat.py,
+    including the round-2 fusions) apply unchanged; only the feature
+    gather/scatter at the boundary is distributed."""
+
+    def __init__(self, dg: DepGraph, device):
+        from .gat import GATLayer
+        from .graph import Chunk
+        self.dg = dg
+        w_csr = dg.edge_weight[dg.csr_from_csc]
+        ch = Chunk(src_s=0, src_e=dg.n_mirrors, dst_s=0, dst_e=dg.owned_n,
+                   column_offset=dg.column_offset,
+                   row_indices=dg.row_indices_m,
+                   edge_weight_forward=dg.edge_weight,
+                   row_offset=dg.row_offset_m,
+                   column_indices=dg.column_indices_l,
+                   edge_weight_backward=np.ascontiguousarray(w_csr))
+        self.layer = GATLayer(ch, dg.n_mirrors, device)
+
+    def forward(self, h_owned, a_src, a_dst, slope=0.2):
+        mirror = dep_nbr_forward(self.dg, h_owned)
+        y, saved = self.layer.forward(mirror, mirror @ a_src,
+                                      h_owned @ a_dst, slope)
+        saved["dist"] = {"mirror": mirror, "a_src": a_src, "a_dst": a_dst}
+        return y, saved
+
+    def backward(self, grad_y, saved, slope=0.2):
+        """Returns the total gradient w.r.t. h_owned."""
+        d = saved["dist"]
+        grad_mirror, g_src_m, g_dst = self.layer.backward(grad_y, saved,
+                                                          slope)
+        grad_mirror = grad_mirror + g_src_m[:, None] * d["a_src"][None, :]
+        grad_h = dep_nbr_backward(self.dg, grad_mirror)
+        grad_h = grad_h + g_dst[:, None] * d["a_dst"][None, :]
+        return grad_h
+
+
 class DistGATLayer:
     """P-partition GAT layer over the dep-neighbor mirror path: forward =
     mirror gather -> per-edge attention -> softmax over the COMPLETE owned
