@@ -163,16 +163,18 @@ int main() {
     NtsVar x = torch::rand({(int64_t)V, 24}, torch::device(dev)) * 2 - 1;
     op::ForwardGPUfuseOp dist(&pg, &active);
     op::ForwardSingleGPUfuseOp single(&pg, &active);
+    /* hub-split atomics reorder fp32 sums between launches:
+     * compare within the north_star tolerance, not bit-equal */
     NtsVar yd = dist.forward(x);
     NtsVar ys = single.forward(x);
-    if (!torch::equal(yd, ys)) {
+    if (!torch::allclose(yd, ys, 1e-4, 1e-5)) {
       fprintf(stderr, "P=1 ForwardGPUfuseOp != ForwardSingleGPUfuseOp\n");
       return 1;
     }
     NtsVar g = torch::rand({(int64_t)V, 24}, torch::device(dev)) * 2 - 1;
     NtsVar gd = dist.backward(g);
     NtsVar gs = single.backward(g);
-    if (!torch::equal(gd, gs)) {
+    if (!torch::allclose(gd, gs, 1e-4, 1e-5)) {
       fprintf(stderr, "P=1 backward mismatch\n");
       return 1;
     }

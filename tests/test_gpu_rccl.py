@@ -95,8 +95,10 @@ def test_dist_fuse_op_world1_matches_single(nccl_world1):
     y_d, y_s = op.forward(x), single.forward(x)
     gx_d, gx_s = op.backward(g), single.backward(g)
     torch.cuda.synchronize()
-    assert torch.equal(y_d, y_s)
-    assert torch.equal(gx_d, gx_s)
+    # hub-split vertices merge partials with fp32 device atomics whose
+    # order varies run to run -> tolerance, not bit-equality
+    assert torch.allclose(y_d, y_s, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(gx_d, gx_s, rtol=1e-4, atol=1e-5)
     # and against the oracle
     ch = host_chunks[0]
     y_ref = oracle.csc_forward(ch.column_offset, ch.row_indices,
