@@ -189,6 +189,27 @@ void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
     const float *lrelu_input, float slope, float *dst_sum,
     const nts_vid *column_offset, nts_vid batch_size, nts_vid feature_size);
 
+/* Fully fused GAT attention forward (f==1 attention scalars): one pass
+ * computes m = s_src_mirror[mirror_index[src]] + s_dst[dst] (stashed to
+ * m_sum_out for the backward's activation mask), leaky-relu(slope), exp,
+ * and per-destination sums; the normalize pass dual-emits the softmax in
+ * CSC (softmax_out) and permuted (softmax_out_perm[perm_pos[e]]) order.
+ * Replaces Scatter_Src_Mirror_to_Msg + Scatter_Dst_to_Msg + elementwise
+ * add/leaky + Edge_Softmax_Forward_Block of the decomposed reference
+ * chain (semantics identical; cache = output, reference convention). */
+void nts_edge_attention_forward(nts_stream *s, float *softmax_out,
+    float *softmax_out_perm, const nts_vid *perm_pos, float *m_sum_out,
+    const float *s_src_mirror, const float *s_dst,
+    const nts_vid *row_indices, const nts_vid *mirror_index, float slope,
+    const nts_vid *column_offset, nts_vid batch_size);
+
+/* Opt-in work-item reuse: with enable=1 the stream caches its two most
+ * recent item decompositions keyed on (offset pointer, batch, edges) and
+ * skips identical rebuilds.  ONLY safe while the caller keeps those
+ * topology buffers live and unchanged (e.g. a layer pinning its chunk);
+ * nts_items_cache_clear drops the cache. */
+void nts_items_reuse(nts_stream *s, int enable);
+
 /* CSR backward gather that ALSO emits the per-edge dot
  * dot_out[dot_pos[e]] = dot(input[column_indices[e]-dst_start], dot_vec[src])
  * from the same streamed row bytes (the GAT attention-scalar gradient).

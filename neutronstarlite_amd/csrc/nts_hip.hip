@@ -70,6 +70,11 @@ struct ItemsBuf {
   uint4 *items = nullptr;
   uint32_t *counter = nullptr;  /* device: n_items after build */
   uint64_t cap = 0;
+  /* opt-in reuse key (see nts_items_reuse): valid only while the caller
+   * guarantees the offset buffer is live and unchanged */
+  const uint32_t *key_off = nullptr;
+  uint32_t key_batch = 0, key_edges = 0;
+  bool valid = false;
 };
 }  // namespace
 
@@ -82,7 +87,12 @@ struct nts_stream {
   std::vector<std::pair<hipEvent_t, hipEvent_t>> freeev;
   double acc_ns[NTS_KTAG_COUNT] = {};
   long long acc_n[NTS_KTAG_COUNT] = {};
-  ItemsBuf items_scratch;  /* per-call work-item buffer (stream-ordered) */
+  ItemsBuf items_scratch[2]; /* work-item buffers (stream-ordered); two
+                                slots so CSC and CSR topologies of one
+                                op-chain can both stay cached under
+                                nts_items_reuse */
+  int items_rr = 0;          /* which slot the next build replaces */
+  bool items_reuse = false;  /* opt-in caching (nts_items_reuse) */
   float *sums_ws = nullptr;  /* per-dst reduction scratch (softmax) */
   uint64_t sums_cap = 0;
 };
@@ -743,6 +753,49 @@ __global__ void k_edge_softmax_norm(const uint4 *__restrict__ items,
   }
 }
 
+/* Fused GAT attention forward, pass 1 (f==1 attention scalars): computes
+ * per edge  m = s_src[mirror_index[src]] + s_dst[dst],  stashes m (the
+ * leaky-relu input the backward mask needs), applies the activation and
+ * exp, and accumulates the per-destination partial sums — replacing the
+ * separate scatter_src_mirror_to_msg + scatter_dst_to_msg kernels and the
+ * torch add/leaky elementwise passes (5 full E-sized passes) with one.
+ * Pass 2 is the shared normalize kernel (dual-order emission). */
+__global__ void k_edge_att_sum(const uint4 *__restrict__ items,
+                               const uint32_t *__restrict__ n_items_p,
+                               float *__restrict__ out /* exp stash */,
+                               float *__restrict__ m_sum_out,
+                               const float *__restrict__ s_src,
+                               const float *__restrict__ s_dst,
+                               const uint32_t *__restrict__ row_indices,
+                               const uint32_t *__restrict__ mirror_index,
+                               float slope, float *__restrict__ sums) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t d = itm.x & 0x7fffffffu;
+    const uint32_t e0 = itm.y;
+    const float sd = s_dst[d];
+    float part = 0.f;
+    for (uint32_t k = lane; k < itm.z; k += 64) {
+      const uint32_t e = e0 + k;
+      const uint32_t src = row_indices[e];
+      const float m = s_src[mirror_index ? mirror_index[src] : src] + sd;
+      m_sum_out[e] = m;
+      const float a = (m > 0.f) ? m : slope * m;
+      const float v = __expf(a); /* no max subtraction: reference semantics,
+                                    ntsCUDADistKernel.cuh:192 */
+      out[e] = v;
+      part += v;
+    }
+#pragma unroll
+    for (int w = 32; w >= 1; w >>= 1) part += __shfl_xor(part, w, 64);
+    if (lane == 0) atomicAdd(&sums[d], part);
+  }
+}
+
 /* per-edge dot: out[e] = dot(dst_rows[d], src_rows[row_indices[e]-src_s]),
  * item-driven (bounded per-wave work even on power-law hubs); lanes stride
  * the feature dim per edge. */
@@ -909,7 +962,20 @@ uint32_t grid_for(uint64_t threads) {
  * gather.  All uses are ordered on the stream, so one scratch suffices. */
 ItemsBuf &get_items(nts_stream *s, const uint32_t *offset, uint32_t batch,
                     uint32_t edges) {
-  ItemsBuf &ib = s->items_scratch;
+  /* Opt-in reuse (nts_items_reuse): when the caller has pinned its
+   * topology, a rebuilt-identical item set is skipped.  Default stays
+   * rebuild-on-every-call — caching by pointer alone is unsound when
+   * chunks are freed and reallocated at the same address. */
+  if (s->items_reuse) {
+    for (auto &slot : s->items_scratch) {
+      if (slot.valid && slot.key_off == offset && slot.key_batch == batch &&
+          slot.key_edges == edges) {
+        return slot;
+      }
+    }
+  }
+  ItemsBuf &ib = s->items_scratch[s->items_rr];
+  s->items_rr ^= 1;
   const uint64_t need = (uint64_t)batch + edges / NTS_SPLIT + 1;
   if (ib.cap < need) {
     NTS_CHECK(hipStreamSynchronize(s->stream));  /* old buffer may be in use */
@@ -926,6 +992,10 @@ ItemsBuf &get_items(nts_stream *s, const uint32_t *offset, uint32_t batch,
                        ib.counter);
   }
   dbg_sync(s, "k_build_items");
+  ib.key_off = offset;
+  ib.key_batch = batch;
+  ib.key_edges = edges;
+  ib.valid = s->items_reuse;
   return ib;
 }
 
@@ -992,8 +1062,10 @@ void nts_stream_destroy(nts_stream *s) {
     hipEventDestroy(p.first);
     hipEventDestroy(p.second);
   }
-  if (s->items_scratch.items) hipFree(s->items_scratch.items);
-  if (s->items_scratch.counter) hipFree(s->items_scratch.counter);
+  for (auto &slot : s->items_scratch) {
+    if (slot.items) hipFree(slot.items);
+    if (slot.counter) hipFree(slot.counter);
+  }
   if (s->sums_ws) hipFree(s->sums_ws);
   if (s->owned) hipStreamDestroy(s->stream);
   delete s;
@@ -1084,8 +1156,13 @@ void nts_gather_by_src_from_dst(nts_stream *s, const float *input,
 }
 
 void nts_items_cache_clear(nts_stream *s) {
-  /* items are rebuilt on every call now; kept for ABI compatibility */
   NTS_CHECK(hipStreamSynchronize(s->stream));
+  for (auto &slot : s->items_scratch) slot.valid = false;
+}
+
+void nts_items_reuse(nts_stream *s, int enable) {
+  if (!enable) nts_items_cache_clear(s);
+  s->items_reuse = enable != 0;
 }
 
 void nts_deserialize_to_gpu(nts_stream *s, float *gpu_buffer, const float *msg,
@@ -1385,6 +1462,33 @@ void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
                       column_offset, batch_size, feature_size,
                       msg_input_grad_perm, perm_pos, lrelu_input, slope,
                       dst_sum);
+}
+
+void nts_edge_attention_forward(nts_stream *s, float *softmax_out,
+                                float *softmax_out_perm,
+                                const nts_vid *perm_pos, float *m_sum_out,
+                                const float *s_src_mirror, const float *s_dst,
+                                const nts_vid *row_indices,
+                                const nts_vid *mirror_index, float slope,
+                                const nts_vid *column_offset,
+                                nts_vid batch_size) {
+  if (!batch_size) return;
+  const uint32_t edges = read_edge_count(s, column_offset, batch_size);
+  if (!edges) return;
+  ItemsBuf &ib = get_items(s, column_offset, batch_size, edges);
+  float *sums = get_sums(s, batch_size);
+  const uint32_t grid =
+      grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
+  Tic t(s, NTS_KTAG_EDGE);
+  hipLaunchKernelGGL(k_edge_att_sum, dim3(grid), dim3(NTS_BLOCK), 0,
+                     s->stream, ib.items, ib.counter, softmax_out, m_sum_out,
+                     s_src_mirror, s_dst, row_indices, mirror_index, slope,
+                     sums);
+  hipLaunchKernelGGL((k_edge_softmax_norm<false>), dim3(grid),
+                     dim3(NTS_BLOCK), 0, s->stream, ib.items, ib.counter,
+                     softmax_out, nullptr, sums, softmax_out_perm, perm_pos,
+                     nullptr, 0.f, nullptr, 1u);
+  dbg_sync(s, "k_edge_att");
 }
 
 int nts_gather_by_src_from_dst_dot(nts_stream *s, const float *input,

@@ -78,37 +78,32 @@ class GATLayer:
         ch, st, E = self.ch, self.stream, self.E
         dev = h.device
         h = h.contiguous()
-        m_src = torch.empty(E, 1, device=dev)
-        m_dst = torch.empty(E, 1, device=dev)
-        st.scatter_src_mirror_to_msg(m_src.data_ptr(),
-                                     s_src.contiguous().data_ptr(),
-                                     ch.row_indices.data_ptr(),
-                                     ch.column_offset.data_ptr(),
-                                     self.mirror_index.data_ptr(),
-                                     ch.dst_n, 1)
-        st.scatter_dst_to_msg(m_dst.data_ptr(), s_dst.contiguous().data_ptr(),
-                              ch.row_indices.data_ptr(),
-                              ch.column_offset.data_ptr(), ch.dst_n, 1)
-        m_sum = m_src + m_dst
-        e = torch.nn.functional.leaky_relu(m_sum, negative_slope)
+        # the layer owns its chunk for its lifetime -> item reuse is safe
+        st.items_reuse(1)
         s = torch.empty(E, 1, device=dev)
         s_csr = torch.empty(E, 1, device=dev)
-        cached = torch.empty(E, 1, device=dev)
-        # dual emission: softmax output lands in CSC order (s) AND directly
-        # in CSR edge order (s_csr) in the same normalize pass, replacing
-        # the backward's nts_permute_f32 (round-1: 2.1 ms/step)
-        st.edge_softmax_forward_dual(s.data_ptr(), s_csr.data_ptr(),
-                                     self._inv_perm_u32.data_ptr(),
-                                     e.contiguous().data_ptr(),
-                                     cached.data_ptr(),
-                                     ch.column_offset.data_ptr(), ch.dst_n, 1)
+        m_sum = torch.empty(E, 1, device=dev)
+        # ONE fused pass computes scatter_src + scatter_dst + leaky_relu +
+        # exp + per-dst sums (5 E-sized passes of the decomposed chain);
+        # the normalize pass dual-emits the softmax in CSC (s) and CSR
+        # (s_csr) edge order.  cache = output (reference convention), so
+        # `s` doubles as the backward's cached tensor.
+        st.edge_attention_forward(s.data_ptr(), s_csr.data_ptr(),
+                                  self._inv_perm_u32.data_ptr(),
+                                  m_sum.data_ptr(),
+                                  s_src.contiguous().data_ptr(),
+                                  s_dst.contiguous().data_ptr(),
+                                  ch.row_indices.data_ptr(),
+                                  self.mirror_index.data_ptr(),
+                                  negative_slope,
+                                  ch.column_offset.data_ptr(), ch.dst_n)
         y = torch.zeros(ch.dst_n, h.shape[1], device=dev)
         st.gather_by_dst_from_src(h.data_ptr(), y.data_ptr(), s.data_ptr(),
                                   ch.row_indices.data_ptr(),
                                   ch.column_offset.data_ptr(),
                                   ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
                                   E, ch.dst_n, h.shape[1], with_weight=True)
-        saved = {"h": h, "s": s, "s_csr": s_csr, "cached": cached,
+        saved = {"h": h, "s": s, "s_csr": s_csr, "cached": s,
                  "m_sum": m_sum}
         return y, saved
 
@@ -119,6 +114,8 @@ class GATLayer:
         dev = grad_y.device
         f = grad_y.shape[1]
         grad_y = grad_y.contiguous()
+        st.items_reuse(1)
+        self.scalar_stream.items_reuse(1)
         # fused CSR gather + edge-dot: grad_h[src] accumulates while the
         # SAME grad_y row bytes produce gs[e] = grad_y[dst(e)].h[src(e)],
         # written straight into CSC slot order (dot_pos = csr->csc map) for

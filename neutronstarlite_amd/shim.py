@@ -93,6 +93,9 @@ def lib():
     l.nts_gather_by_src_from_dst_dot.argtypes = (
         [_vp] + [_vp] * 5 + [_u32] * 4 + [_vp] * 3)
     l.nts_gather_by_src_from_dst_dot.restype = _i32
+    l.nts_edge_attention_forward.argtypes = (
+        [_vp] + [_vp] * 8 + [_c.c_float] + [_vp] + [_u32])
+    l.nts_items_reuse.argtypes = [_vp, _i32]
     l.nts_permute_f32.argtypes = [_vp, _vp, _vp, _vp, _i64]
     l.nts_sample_reservoir.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32,
                                        _c.c_ulonglong, _vp, _vp]
@@ -289,6 +292,18 @@ class Stream:
 
     def items_cache_clear(self):
         self._lib.nts_items_cache_clear(self.h)
+
+    def items_reuse(self, enable):
+        """Opt-in work-item caching; caller must pin its topology buffers."""
+        self._lib.nts_items_reuse(self.h, 1 if enable else 0)
+
+    def edge_attention_forward(self, softmax_out, softmax_out_perm, perm_pos,
+                               m_sum_out, s_src_mirror, s_dst, row_indices,
+                               mirror_index, slope, column_offset, batch):
+        self._lib.nts_edge_attention_forward(
+            self.h, _vp(softmax_out), _vp(softmax_out_perm), _vp(perm_pos),
+            _vp(m_sum_out), _vp(s_src_mirror), _vp(s_dst), _vp(row_indices),
+            _vp(mirror_index), slope, _vp(column_offset), batch)
 
     def destroy(self):
         if self.h:
