@@ -331,6 +331,110 @@ __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm(
  * slot map, so the softmax backward consumes it without a permute pass).
  * Clean-case kernel: f % ELEM == 0 and f <= G*ELEM (one slab) — the
  * launcher falls back to the unfused pair otherwise. */
+/* Batched-LDS variant of the per-edge dot reduction: instead of a
+ * log2(G)-step shuffle chain per edge (ds_permute latency on every edge),
+ * lanes stash their per-edge partials for a batch of NTS_DOT_B edges into
+ * padded LDS, then NTS_DOT_B lanes each sum one edge's G partials with
+ * conflict-free stride-(G+1) reads — ~2 LDS ops per lane per edge and the
+ * reads pipeline.  The gather accumulation is identical to
+ * k_gather_spmm_dot (registers, RMW/atomic store). */
+constexpr uint32_t NTS_DOT_B = 16; /* edges per LDS transpose batch */
+
+template <int ELEM>
+__global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm_dot_lds(
+    const uint4 *__restrict__ items, const uint32_t *__restrict__ n_items_p,
+    const uint32_t *__restrict__ nbr, const float *__restrict__ ew,
+    const float *__restrict__ in, float *__restrict__ out, uint32_t nbr_start,
+    const float *__restrict__ dot_vec, float *__restrict__ dot_out,
+    const uint32_t *__restrict__ dot_pos, uint32_t f, uint32_t G) {
+  /* per-group LDS: NTS_DOT_B rows of (G+1) floats (pad kills bank
+   * conflicts); groups tile the block back to back.  Sized for the worst
+   * case G=16 (most groups per block: 16 x 16x17 floats). */
+  __shared__ float lds[NTS_DOT_B * (16 + 1) * (NTS_BLOCK / 16)];
+  const uint32_t n_items = *n_items_p;
+  const uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const uint32_t glane = threadIdx.x & (G - 1);
+  const uint32_t n_groups = (gridDim.x * blockDim.x) / G;
+  float *gl = lds + (threadIdx.x / G) * NTS_DOT_B * (G + 1);
+  for (uint32_t it = tid / G; it < n_items; it += n_groups) {
+    const uint4 itm = items[it];
+    const uint32_t v = itm.x & 0x7fffffffu;
+    const bool shared_v = (itm.x >> 31) != 0;
+    const uint32_t e0 = itm.y, cnt = itm.z;
+    const uint32_t off = glane * ELEM;
+    const bool act = off + ELEM <= f;
+    float acc[ELEM] = {};
+    float hreg[ELEM] = {};
+    if (act) {
+      const float *hp = dot_vec + (uint64_t)v * f + off;
+#pragma unroll
+      for (int j = 0; j < ELEM; ++j) hreg[j] = hp[j];
+    }
+    for (uint32_t b = 0; b < cnt; b += NTS_DOT_B) {
+      const uint32_t nb = min(NTS_DOT_B, cnt - b);
+      uint32_t k = 0;
+      for (; k + 4 <= nb; k += 4) {
+        const uint32_t e = e0 + b + k;
+        float dp[4] = {};
+        if (act) {
+          const float *p[4];
+          float wv[4];
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            p[q] = in + (uint64_t)(nbr[e + q] - nbr_start) * f + off;
+            wv[q] = ew[e + q];
+          }
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+#pragma unroll
+            for (int j = 0; j < ELEM; ++j) {
+              const float x = p[q][j];
+              acc[j] = fmaf(wv[q], x, acc[j]);
+              dp[q] = fmaf(hreg[j], x, dp[q]);
+            }
+          }
+        }
+#pragma unroll
+        for (int q = 0; q < 4; ++q) gl[(k + q) * (G + 1) + glane] = dp[q];
+      }
+      for (; k < nb; ++k) {
+        const uint32_t e = e0 + b + k;
+        float dp = 0.f;
+        if (act) {
+          const float w = ew[e];
+          const float *p = in + (uint64_t)(nbr[e] - nbr_start) * f + off;
+#pragma unroll
+          for (int j = 0; j < ELEM; ++j) {
+            const float x = p[j];
+            acc[j] = fmaf(w, x, acc[j]);
+            dp = fmaf(hreg[j], x, dp);
+          }
+        }
+        gl[k * (G + 1) + glane] = dp;
+      }
+      __builtin_amdgcn_wave_barrier(); /* group lives in one wave (G<=64) */
+      if (glane < nb) {
+        float sum = 0.f;
+        const float *row = gl + glane * (G + 1);
+        for (uint32_t j = 0; j < G; ++j) sum += row[j];
+        const uint32_t e = e0 + b + glane;
+        dot_out[dot_pos ? dot_pos[e] : e] = sum;
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+    if (act) {
+      float *o = out + (uint64_t)v * f + off;
+      if (!shared_v) {
+#pragma unroll
+        for (int j = 0; j < ELEM; ++j) o[j] += acc[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < ELEM; ++j) atomicAdd(&o[j], acc[j]);
+      }
+    }
+  }
+}
+
 template <int ELEM>
 __global__ __launch_bounds__(NTS_BLOCK) void k_gather_spmm_dot(
     const uint4 *__restrict__ items, const uint32_t *__restrict__ n_items_p,
@@ -1523,18 +1627,23 @@ int nts_gather_by_src_from_dst_dot(nts_stream *s, const float *input,
   ItemsBuf &ib = get_items(s, row_offset, batch_size, edges);
   const uint64_t bound_groups = (uint64_t)batch_size + edges / NTS_SPLIT + 1;
   const uint32_t grid = grid_for(bound_groups * G);
+  /* NTS_GATHER_DOT: 1 = batched LDS-transpose reduction (default),
+   * 2 = per-edge shuffle chains (kept selectable for A/B). */
+  static const uint32_t variant = env_u32("NTS_GATHER_DOT", 1);
   Tic t(s, NTS_KTAG_BWD);
-  if (elem == 4) {
-    hipLaunchKernelGGL((k_gather_spmm_dot<4>), dim3(grid), dim3(NTS_BLOCK), 0,
-                       s->stream, ib.items, ib.counter, column_indices,
-                       weight_backward, input, output, dst_start, dot_vec,
-                       dot_out, dot_pos, f, G);
+#define NTS_DOT_LAUNCH(K, E)                                                  \
+  hipLaunchKernelGGL((K<E>), dim3(grid), dim3(NTS_BLOCK), 0, s->stream,       \
+                     ib.items, ib.counter, column_indices, weight_backward,   \
+                     input, output, dst_start, dot_vec, dot_out, dot_pos, f,  \
+                     G)
+  if (variant == 2) {
+    if (elem == 4) NTS_DOT_LAUNCH(k_gather_spmm_dot, 4);
+    else NTS_DOT_LAUNCH(k_gather_spmm_dot, 2);
   } else {
-    hipLaunchKernelGGL((k_gather_spmm_dot<2>), dim3(grid), dim3(NTS_BLOCK), 0,
-                       s->stream, ib.items, ib.counter, column_indices,
-                       weight_backward, input, output, dst_start, dot_vec,
-                       dot_out, dot_pos, f, G);
+    if (elem == 4) NTS_DOT_LAUNCH(k_gather_spmm_dot_lds, 4);
+    else NTS_DOT_LAUNCH(k_gather_spmm_dot_lds, 2);
   }
+#undef NTS_DOT_LAUNCH
   dbg_sync(s, "k_gather_spmm_dot");
   return 1;
 }
