@@ -137,3 +137,64 @@ def test_edge_softmax_fwd_bwd(setup, f):
                              setup["d_coff"].data_ptr(), ch.dst_n, f)
     torch.cuda.synchronize()
     assert_close(gin, gin_ref, "softmax bwd")
+
+
+def test_fused_attention_forward_matches_decomposed(setup):
+    """nts_edge_attention_forward (one pass for scatter_src + scatter_dst +
+    leaky_relu + exp + sums, dual-order softmax emission) vs the decomposed
+    kernel chain it replaces, and nts_edge_softmax_forward_dual vs the plain
+    softmax + permute."""
+    st, ch, dev = setup["s"], setup["ch"], setup["dev"]
+    E, M = ch.edge_size, len(setup["uniq"])
+    slope = 0.2
+    rng = np.random.default_rng(9)
+    s_src = torch.from_numpy(rng.normal(size=(M, 1)).astype(np.float32)).to(dev)
+    s_dst = torch.from_numpy(
+        rng.normal(size=(ch.dst_n, 1)).astype(np.float32)).to(dev)
+    # decomposed chain
+    m_src = torch.empty(E, 1, device=dev)
+    m_dst = torch.empty(E, 1, device=dev)
+    st.scatter_src_mirror_to_msg(m_src.data_ptr(), s_src.data_ptr(),
+                                 setup["d_rows"].data_ptr(),
+                                 setup["d_coff"].data_ptr(),
+                                 setup["d_mi"].data_ptr(), ch.dst_n, 1)
+    st.scatter_dst_to_msg(m_dst.data_ptr(), s_dst.data_ptr(),
+                          setup["d_rows"].data_ptr(),
+                          setup["d_coff"].data_ptr(), ch.dst_n, 1)
+    m_sum_ref = m_src + m_dst
+    e_val = torch.nn.functional.leaky_relu(m_sum_ref, slope).contiguous()
+    s_ref = torch.empty(E, 1, device=dev)
+    cached_ref = torch.empty(E, 1, device=dev)
+    st.edge_softmax_forward(s_ref.data_ptr(), e_val.data_ptr(),
+                            cached_ref.data_ptr(), setup["d_rows"].data_ptr(),
+                            setup["d_coff"].data_ptr(), ch.dst_n, 1)
+    # permutation map (CSC -> CSR)
+    perm = np.argsort(ch.row_indices.astype(np.int64), kind="stable")
+    inv = np.empty_like(perm)
+    inv[perm] = np.arange(len(perm))
+    d_inv = torch.from_numpy(inv.astype(np.uint32).view(np.int32)).to(dev)
+    # fused
+    s_f = torch.empty(E, 1, device=dev)
+    s_f_csr = torch.empty(E, 1, device=dev)
+    m_sum_f = torch.empty(E, 1, device=dev)
+    st.edge_attention_forward(s_f.data_ptr(), s_f_csr.data_ptr(),
+                              d_inv.data_ptr(), m_sum_f.data_ptr(),
+                              s_src.data_ptr(), s_dst.data_ptr(),
+                              setup["d_rows"].data_ptr(),
+                              setup["d_mi"].data_ptr(), slope,
+                              setup["d_coff"].data_ptr(), ch.dst_n)
+    torch.cuda.synchronize()
+    assert_close(m_sum_f, m_sum_ref.cpu().numpy(), "fused m_sum")
+    assert_close(s_f, s_ref.cpu().numpy(), "fused softmax")
+    assert_close(s_f_csr, s_ref.cpu().numpy()[perm], "fused softmax CSR order")
+    # the dual-emitting softmax entry alone
+    s_d = torch.empty(E, 1, device=dev)
+    s_d_csr = torch.empty(E, 1, device=dev)
+    cc = torch.empty(E, 1, device=dev)
+    st.edge_softmax_forward_dual(s_d.data_ptr(), s_d_csr.data_ptr(),
+                                 d_inv.data_ptr(), e_val.data_ptr(),
+                                 cc.data_ptr(), setup["d_coff"].data_ptr(),
+                                 ch.dst_n, 1)
+    torch.cuda.synchronize()
+    assert_close(s_d, s_ref.cpu().numpy(), "dual softmax")
+    assert_close(s_d_csr, s_ref.cpu().numpy()[perm], "dual softmax CSR")
