@@ -203,6 +203,13 @@ void nts_edge_attention_forward(nts_stream *s, float *softmax_out,
     const nts_vid *row_indices, const nts_vid *mirror_index, float slope,
     const nts_vid *column_offset, nts_vid batch_size);
 
+/* Per-vertex sum of per-edge scalars over an offset array (CSC or CSR):
+ * out[v] += sum of weights[e] over v's edges — the f=1 degenerate of the
+ * weighted gather, done at full lane occupancy (the GAT attention-scalar
+ * source gradient).  out is caller-zeroed. */
+void nts_weight_sum(nts_stream *s, float *out, const float *weights,
+    const nts_vid *offset, nts_vid batch_size);
+
 /* Opt-in work-item reuse: with enable=1 the stream caches its two most
  * recent item decompositions keyed on (offset pointer, batch, edges) and
  * skips identical rebuilds.  ONLY safe while the caller keeps those

@@ -900,6 +900,36 @@ __global__ void k_edge_att_sum(const uint4 *__restrict__ items,
   }
 }
 
+/* per-vertex sum of per-edge scalars: out[v] = sum_{e in item(v)} w[e] —
+ * the f=1 "aggregate an all-ones input with weights w" reduction (GAT's
+ * attention-scalar source gradient) done natively: lanes stride the item's
+ * edges and wave-reduce, instead of the f-wide gather kernel where 15/16
+ * lanes idle at f=1 (measured 1.9 ms -> this shape streams w + offsets
+ * only).  Hub items merge by fp32 atomics like the gather. */
+__global__ void k_weight_sum(const uint4 *__restrict__ items,
+                             const uint32_t *__restrict__ n_items_p,
+                             float *__restrict__ out,
+                             const float *__restrict__ w) {
+  const uint32_t n_items = *n_items_p;
+  const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t n_waves = (gridDim.x * blockDim.x) >> 6;
+  for (uint32_t it = wave; it < n_items; it += n_waves) {
+    const uint4 itm = items[it];
+    const uint32_t v = itm.x & 0x7fffffffu;
+    const bool shared_v = (itm.x >> 31) != 0;
+    const uint32_t e0 = itm.y;
+    float part = 0.f;
+    for (uint32_t k = lane; k < itm.z; k += 64) part += w[e0 + k];
+#pragma unroll
+    for (int q = 32; q >= 1; q >>= 1) part += __shfl_xor(part, q, 64);
+    if (lane == 0) {
+      if (shared_v) atomicAdd(&out[v], part);
+      else out[v] += part;
+    }
+  }
+}
+
 /* per-edge dot: out[e] = dot(dst_rows[d], src_rows[row_indices[e]-src_s]),
  * item-driven (bounded per-wave work even on power-law hubs); lanes stride
  * the feature dim per edge. */
@@ -1566,6 +1596,20 @@ void nts_edge_softmax_backward_fused(nts_stream *s, float *msg_input_grad,
                       column_offset, batch_size, feature_size,
                       msg_input_grad_perm, perm_pos, lrelu_input, slope,
                       dst_sum);
+}
+
+void nts_weight_sum(nts_stream *s, float *out, const float *weights,
+                    const nts_vid *offset, nts_vid batch_size) {
+  if (!batch_size) return;
+  const uint32_t edges = read_edge_count(s, offset, batch_size);
+  if (!edges) return;
+  ItemsBuf &ib = get_items(s, offset, batch_size, edges);
+  const uint32_t grid =
+      grid_for(((uint64_t)batch_size + edges / NTS_SPLIT) * 64);
+  Tic t(s, NTS_KTAG_EDGE);
+  hipLaunchKernelGGL(k_weight_sum, dim3(grid), dim3(NTS_BLOCK), 0, s->stream,
+                     ib.items, ib.counter, out, weights);
+  dbg_sync(s, "k_weight_sum");
 }
 
 void nts_edge_attention_forward(nts_stream *s, float *softmax_out,

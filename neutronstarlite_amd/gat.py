@@ -153,18 +153,13 @@ class GATLayer:
         # a 115 ms step), while the gather's work items split hubs and merge
         # a handful of partials.
         # g_dst came out of the softmax-backward pass itself (dst_sum);
-        # g_src still needs the src-major reduction — through the
-        # load-balanced gather kernel (per-edge values as WEIGHTS over an
-        # all-ones input): direct msg->vertex atomics serialize on
-        # power-law hub sources (one address takes every hub edge's
-        # atomicAdd — measured 25 ms of a 115 ms step in round 1).
-        if self._ones_dst is None:
-            self._ones_dst = torch.ones(ch.dst_n, 1, device=dev)
+        # g_src is the src-major sum of ge over the CSR — the dedicated
+        # weight-sum kernel (full lane occupancy at f=1; the f-wide gather
+        # left 15/16 lanes idle, and direct msg->vertex atomics serialize
+        # on power-law hub sources — measured 25 ms of a 115 ms step in
+        # round 1).  Item splitting keeps hubs bounded here too.
         sst = self.scalar_stream
         g_src = torch.zeros(ch.src_n, 1, device=dev)
-        sst.gather_by_src_from_dst(self._ones_dst.data_ptr(), g_src.data_ptr(),
-                                  ge_csr.data_ptr(), ch.row_offset.data_ptr(),
-                                  ch.column_indices.data_ptr(),
-                                  ch.src_s, ch.src_e, ch.dst_s, ch.dst_e,
-                                  E, ch.src_n, 1, with_weight=True)
+        sst.weight_sum(g_src.data_ptr(), ge_csr.data_ptr(),
+                       ch.row_offset.data_ptr(), ch.src_n)
         return grad_h, g_src[:, 0], g_dst[:, 0]

@@ -96,6 +96,7 @@ def lib():
     l.nts_edge_attention_forward.argtypes = (
         [_vp] + [_vp] * 8 + [_c.c_float] + [_vp] + [_u32])
     l.nts_items_reuse.argtypes = [_vp, _i32]
+    l.nts_weight_sum.argtypes = [_vp, _vp, _vp, _vp, _u32]
     l.nts_permute_f32.argtypes = [_vp, _vp, _vp, _vp, _i64]
     l.nts_sample_reservoir.argtypes = [_vp, _vp, _vp, _vp, _u32, _u32,
                                        _c.c_ulonglong, _vp, _vp]
@@ -296,6 +297,10 @@ class Stream:
     def items_reuse(self, enable):
         """Opt-in work-item caching; caller must pin its topology buffers."""
         self._lib.nts_items_reuse(self.h, 1 if enable else 0)
+
+    def weight_sum(self, out, weights, offset, batch):
+        self._lib.nts_weight_sum(self.h, _vp(out), _vp(weights), _vp(offset),
+                                 batch)
 
     def edge_attention_forward(self, softmax_out, softmax_out_perm, perm_pos,
                                m_sum_out, s_src_mirror, s_dst, row_indices,
