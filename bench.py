@@ -283,6 +283,11 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         dchunks = chunks
         engine = _test_engine_factory()
     engine.stream.timing(True)
+    if _test_engine_factory is None and args.model != "gcn-sample":
+        # chunks are static for the full-batch modes -> work-item reuse is
+        # safe (sampled subgraphs change every step, so gcn-sample keeps
+        # the rebuild-per-call default)
+        engine.stream.items_reuse(1)
     rg = RingGraph(offs, rank, dchunks, dev)
     if distributed and args.mirror_filtered:
         from neutronstarlite_amd.ring import setup_mirror_lists
