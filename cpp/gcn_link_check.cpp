@@ -28,18 +28,18 @@
 
 using namespace nts;
 
-/* ---- the vendored loop shape (GCN.hpp:217-235) ---- */
+/* ---- the vendored loop shapes (GCN.hpp:205-235) ---- */
 struct GCNLoop {
   PartitionedGraph *partitioned_graph;
   VertexSubset *active;
   NtsContext *ctx;
-  std::vector<NtsVar> X; /* X[0..layers] as in GCN.hpp:108-117 */
-  std::vector<NtsVar> W;
+  std::vector<NtsVar> X;          /* X[0..layers] as in GCN.hpp:108-117 */
+  std::vector<Parameter *> P;     /* layer weights, GCN.hpp:96-104 */
 
   NtsVar vertexForward(NtsVar &a, NtsVar &x, int layer) {
     (void)x; /* GCN.hpp:183-195: second input unused by the GCN model */
-    if (layer == 0) return torch::relu(torch::mm(a, W[0]));
-    return torch::log_softmax(torch::mm(a, W[1]), 1);
+    if (layer == 0) return torch::relu(P[0]->forward(a));
+    return torch::log_softmax(P[1]->forward(a), 1);
   }
 
   void Forward(int layers) {
@@ -52,6 +52,16 @@ struct GCNLoop {
             return vertexForward(n_i, v_i, i);
           },
           Y_i, X[i]);
+    }
+  }
+
+  void Update() {
+    /* GCN.hpp:205-214, names unchanged (grad stays on device — the
+     * reference's .cpu() bounce is not required by the surface) */
+    for (size_t i = 0; i < P.size(); i++) {
+      P[i]->all_reduce_to_gradient(P[i]->W.grad());
+      P[i]->learnC2G_with_decay_Adam();
+      P[i]->next();
     }
   }
 };
@@ -127,7 +137,9 @@ int main() {
     printf("rccl grouped self send/recv ok\n");
   }
 
-  /* the flagship loop, world 1 */
+  /* the flagship loop, world 1: Forward (GCN.hpp:217-235) +
+   * self_backward + Update (GCN.hpp:205-214 over Parameter/RCCL) for a
+   * few epochs; the loss must drop */
   GCNLoop loop;
   loop.partitioned_graph = &pg;
   loop.active = &active;
@@ -135,28 +147,31 @@ int main() {
   loop.ctx = &ctx;
   loop.X.resize(3);
   loop.X[0] = torch::rand({(int64_t)V, F0}, torch::device(dev)) * 2 - 1;
-  loop.W.push_back(torch::empty({F0, F1}, torch::device(dev))
-                       .uniform_(-0.1, 0.1).set_requires_grad(true));
-  loop.W.push_back(torch::empty({F1, C}, torch::device(dev))
-                       .uniform_(-0.1, 0.1).set_requires_grad(true));
-  loop.Forward(2);
+  Parameter P0(F0, F1, 1e-2f), P1(F1, C, 1e-2f);
+  for (Parameter *p : {&P0, &P1}) {
+    p->comm = comm;
+    p->stream = pg.stream;
+    p->to(dev);
+    p->init_parameter();  /* rank-0 bcast over RCCL */
+    loop.P.push_back(p);
+  }
   NtsVar labels = torch::randint(C, {(int64_t)V},
                                  torch::device(dev).dtype(torch::kLong));
-  NtsVar loss = torch::nll_loss(loop.X[2], labels);
-  ctx.self_backward(loss);
-  printf("flagship loop fwd+bwd ok, loss %.4f\n", loss.item<double>());
-
-  /* weight-grad allreduce + init bcast (Update()'s comm, world 1) */
-  for (auto &Wl : loop.W) {
-    NtsVar g = Wl.grad();
-    if (!g.defined()) { fprintf(stderr, "missing weight grad\n"); return 1; }
-    NtsVar gc = g.contiguous();
-    nts_comm_allreduce_sum_f32(comm, pg.stream, gc.data_ptr<float>(),
-                               gc.data_ptr<float>(), gc.numel());
-    nts_comm_bcast_f32(comm, pg.stream, Wl.data_ptr<float>(), Wl.numel(), 0);
+  double first_loss = 0, last_loss = 0;
+  for (int ep = 0; ep < 8; ep++) {
+    loop.Forward(2);
+    NtsVar loss = torch::nll_loss(loop.X[2], labels);
+    ctx.self_backward(loss);
+    loop.Update();
+    last_loss = loss.item<double>();
+    if (ep == 0) first_loss = loss.item<double>();
   }
-  nts_stream_sync(pg.stream);
-  printf("rccl allreduce+bcast ok\n");
+  printf("flagship loop fwd+bwd+Update ok, loss %.4f -> %.4f\n", first_loss,
+         last_loss);
+  if (!(last_loss < first_loss)) {
+    fprintf(stderr, "flagship loop: NO LEARNING\n");
+    return 1;
+  }
 
   /* P=1 equivalence: ForwardGPUfuseOp must match ForwardSingleGPUfuseOp */
   {

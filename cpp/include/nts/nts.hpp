@@ -434,6 +434,101 @@ class DistGPUEdgeSoftMax : public ntsGraphOp {
 
 }  // namespace op
 
+/* Layer weights + hand-rolled Adam + gradient allreduce (Parameter
+ * surface, core/NtsScheduler.hpp:639-791), so toolkits' Update() loops —
+ * P[i]->all_reduce_to_gradient(...); P[i]->learnC2G_with_decay_Adam();
+ * P[i]->next(); (toolkits/GCN.hpp:205-214) — compile unchanged.
+ *
+ * MI355X-native differences (surface identical, plumbing new): the
+ * reference allreduces on HOST through Network_simple's MPI_Allreduce
+ * (network.h:198-203) and keeps Adam state split CPU/GPU; here gradients
+ * and Adam state stay in HBM and the allreduce is RCCL over xGMI
+ * (nts_comm_allreduce_sum_f32) on the caller's stream — comm == nullptr
+ * degenerates to single-rank (no-op allreduce), matching np=1. */
+struct Parameter {
+  NtsVar W;       /* weights (leaf, requires_grad) */
+  NtsVar W_gradient, M_GPU, V_GPU, W_g;
+  ValueType alpha = 0.01f, beta1 = 0.9f, beta2 = 0.999f, epsilon = 1e-9f;
+  ValueType alpha_t, beta1_t, beta2_t;
+  ValueType weight_decay = 0.f, decay_rate = 1.f;
+  long decay_epoch = -1, curr_epoch = 0;
+  long row, col;
+  nts_comm *comm = nullptr;
+  nts_stream *stream = nullptr;
+
+  /* Xavier-uniform init, NtsScheduler.hpp:669-672 convention */
+  Parameter(size_t w, size_t h, ValueType alpha_ = 0.01f,
+            ValueType beta1_ = 0.9f, ValueType beta2_ = 0.999f,
+            ValueType epsilon_ = 1e-9f, ValueType weight_decay_ = 0.f)
+      : alpha(alpha_), beta1(beta1_), beta2(beta2_), epsilon(epsilon_),
+        weight_decay(weight_decay_), row((long)w), col((long)h) {
+    const ValueType scale = std::sqrt(6.0 / (w + h));
+    W = ((2 * scale) * torch::rand({(long)w, (long)h}) - scale)
+            .set_requires_grad(true);
+    alpha_t = alpha; beta1_t = beta1; beta2_t = beta2;
+  }
+
+  void to(torch::Device dev) {
+    W = W.detach().to(dev).set_requires_grad(true);
+    Adam_to_GPU(dev);
+  }
+  void Adam_to_GPU(torch::Device dev) { /* NtsScheduler.hpp:754-758 */
+    M_GPU = torch::zeros({row, col}, torch::device(dev));
+    V_GPU = torch::zeros({row, col}, torch::device(dev));
+  }
+
+  /* rank-0 weight broadcast at init (NtsScheduler.hpp:716-718 ->
+   * MPI_Bcast; here ncclBroadcast) */
+  void init_parameter() {
+    if (comm && W.is_cuda()) {
+      NtsVar w = W.detach().contiguous();
+      nts_comm_bcast_f32(comm, stream, w.data_ptr<float>(), w.numel(), 0);
+    }
+  }
+
+  /* DDP gradient sum (NtsScheduler.hpp:719-722); accepts the grad on any
+   * device — no .cpu() bounce is required (or useful) here */
+  void all_reduce_to_gradient(NtsVar from) {
+    W_gradient = from.contiguous();
+    if (comm && W_gradient.is_cuda()) {
+      nts_comm_allreduce_sum_f32(comm, stream, W_gradient.data_ptr<float>(),
+                                 W_gradient.data_ptr<float>(),
+                                 W_gradient.numel());
+      nts_stream_sync(stream);
+    }
+  }
+
+  void set_decay(ValueType decay_rate_, long decay_epoch_) {
+    decay_rate = decay_rate_;
+    decay_epoch = decay_epoch_;
+  }
+
+  /* epoch bookkeeping; keeps the reference's (idiosyncratic) running
+   * bias-correction exactly (NtsScheduler.hpp:725-733) */
+  void next() {
+    if (decay_epoch != -1 && curr_epoch != 0 &&
+        curr_epoch % decay_epoch == 0)
+      alpha_t *= decay_rate;
+    alpha = alpha_t * std::sqrt(1 - beta2) / (1 - beta1);
+    beta1 *= beta1_t;
+    beta2 *= beta2_t;
+    curr_epoch++;
+  }
+
+  NtsVar forward(NtsVar x) { return x.mm(W); }
+
+  /* Adam step with decoupled weight decay (NtsScheduler.hpp:759-767
+   * semantics), entirely on device */
+  void learnC2G_with_decay_Adam() {
+    torch::NoGradGuard ng;
+    NtsVar g = W_gradient.to(W.device()) + weight_decay * W.detach();
+    M_GPU = beta1 * M_GPU + (1 - beta1) * g;
+    V_GPU = beta2 * V_GPU + (1 - beta2) * g * g;
+    W.set_data(W.detach() - alpha * M_GPU / (torch::sqrt(V_GPU) + epsilon));
+    if (W.grad().defined()) W.mutable_grad().zero_();
+  }
+};
+
 /* Tape-based context (NtsContext surface, core/ntsContext.hpp:108-359):
  * graph ops bypass libtorch autograd; NN segments use it.  self_backward
  * walks the tape mixing torch::autograd::grad with ntsGraphOp::backward. */
