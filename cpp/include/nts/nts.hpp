@@ -90,6 +90,7 @@ struct PartitionedGraph {
   int partition_id = 0;
   std::vector<VertexId> partition_offset;  // [P+1]
   nts_stream *stream = nullptr;            // compute stream (C-ABI)
+  nts_stream *comm_stream = nullptr;       // ring-exchange stream (overlap)
   nts_comm *comm = nullptr;                // RCCL ring (nullptr => P==1 only)
   torch::Tensor mirror_index;              // per-graph mirror compression
                                            // (generateMirrorIndex surface,
@@ -97,9 +98,13 @@ struct PartitionedGraph {
 
   /* Wrap the HIP null stream (= torch's default stream): libtorch tensor
    * fills/copies and our kernels must share one stream order. */
-  PartitionedGraph() { stream = nts_stream_wrap(nullptr); }
+  PartitionedGraph() {
+    stream = nts_stream_wrap(nullptr);
+    comm_stream = nts_stream_create();  /* non-blocking side stream */
+  }
   ~PartitionedGraph() {
     if (stream) nts_stream_destroy(stream);
+    if (comm_stream) nts_stream_destroy(comm_stream);
   }
   VertexId owned_vertices() const {
     return partition_offset[partition_id + 1] - partition_offset[partition_id];
@@ -230,17 +235,37 @@ class ForwardGPUfuseOp : public ntsGraphOp {
     assert(P == 1 || pg->comm);
     NtsVar x = f_input.contiguous();
     NtsVar y = torch::zeros({(int64_t)pg->owned_vertices(), f}, x.options());
-    detail::csc_forward(pg->stream, *subgraphs[r], x, y);
-    for (int step = 1; step < P; step++) {
+    /* overlapped ring (PROC_OVERLAP default, graph.hpp:3490-3535): the
+     * exchange for step s+1 runs on comm_stream while step s's block
+     * aggregates on the compute stream.  Buffer safety across streams
+     * relies on the wait points below: a tensor freed on the compute
+     * stream is only REUSED by compute-stream work, which the waits order
+     * after the comm stream finished touching it. */
+    auto post = [&](int step, NtsVar &recv) {
       const int to = (r - step + P) % P, frm = (r + step) % P;
-      NtsVar recv = torch::empty({(int64_t)pg->part_n(frm), f}, x.options());
+      recv = torch::empty({(int64_t)pg->part_n(frm), f}, x.options());
       nts_comm_group_begin();
-      nts_comm_send_f32(pg->comm, pg->stream, x.data_ptr<float>(),
+      nts_comm_send_f32(pg->comm, pg->comm_stream, x.data_ptr<float>(),
                         x.numel(), to);
-      nts_comm_recv_f32(pg->comm, pg->stream, recv.data_ptr<float>(),
+      nts_comm_recv_f32(pg->comm, pg->comm_stream, recv.data_ptr<float>(),
                         recv.numel(), frm);
       nts_comm_group_end();
-      detail::csc_forward(pg->stream, *subgraphs[frm], recv, y);
+    };
+    NtsVar bufs[2];
+    if (P > 1) {
+      /* x is produced on the compute stream; one dependency covers every
+       * later send of the same buffer */
+      nts_stream_wait_stream(pg->comm_stream, pg->stream);
+      post(1, bufs[1 % 2]);
+    }
+    detail::csc_forward(pg->stream, *subgraphs[r], x, y);
+    for (int step = 1; step < P; step++) {
+      /* comm tail == exchange `step` here (step+1 not yet posted), so
+       * this wait releases exactly when our block has arrived */
+      nts_stream_wait_stream(pg->stream, pg->comm_stream);
+      if (step + 1 < P) post(step + 1, bufs[(step + 1) % 2]);
+      detail::csc_forward(pg->stream, *subgraphs[(r + step) % P],
+                          bufs[step % 2], y);
     }
     return y;
   }
@@ -253,19 +278,35 @@ class ForwardGPUfuseOp : public ntsGraphOp {
     NtsVar g = output_grad.contiguous();
     NtsVar gx = torch::zeros({(int64_t)pg->owned_vertices(), f}, g.options());
     detail::csr_backward(pg->stream, *subgraphs[r], g, gx);
+    /* pipelined: step s+1's partial computes while step s's exchange is
+     * in flight (compute_sync_decoupled semantics, graph.hpp:3456-3622).
+     * Per iteration: compute partial_s -> merge recv_{s-1} (the comm tail
+     * is exchange s-1 at that point, so the wait releases exactly on its
+     * arrival) -> post exchange_s. */
+    NtsVar recvs[2], partials[2];
     for (int step = 1; step < P; step++) {
-      const int k = (r + step) % P;          /* owner we feed */
+      const int k = (r + step) % P;            /* owner we feed */
       const int peer_src = (r - step + P) % P; /* partial arriving for us */
-      NtsVar partial = torch::zeros({(int64_t)pg->part_n(k), f}, g.options());
+      NtsVar &partial = partials[step % 2];
+      partial = torch::zeros({(int64_t)pg->part_n(k), f}, g.options());
       detail::csr_backward(pg->stream, *subgraphs[k], g, partial);
-      NtsVar recv = torch::empty_like(gx);
+      if (step >= 2) {
+        nts_stream_wait_stream(pg->stream, pg->comm_stream);
+        gx += recvs[(step - 1) % 2];
+      }
+      nts_stream_wait_stream(pg->comm_stream, pg->stream); /* partial done */
+      NtsVar &recv = recvs[step % 2];
+      recv = torch::empty_like(gx);
       nts_comm_group_begin();
-      nts_comm_send_f32(pg->comm, pg->stream, partial.data_ptr<float>(),
+      nts_comm_send_f32(pg->comm, pg->comm_stream, partial.data_ptr<float>(),
                         partial.numel(), k);
-      nts_comm_recv_f32(pg->comm, pg->stream, recv.data_ptr<float>(),
+      nts_comm_recv_f32(pg->comm, pg->comm_stream, recv.data_ptr<float>(),
                         recv.numel(), peer_src);
       nts_comm_group_end();
-      gx += recv;
+    }
+    if (P > 1) {
+      nts_stream_wait_stream(pg->stream, pg->comm_stream);
+      gx += recvs[(P - 1) % 2];
     }
     return gx;
   }

@@ -39,6 +39,11 @@ nts_stream *nts_stream_wrap(void *hip_stream);       /* wrap an externally owned
 void nts_stream_destroy(nts_stream *s);              /* Cuda_Stream::destory_Stream */
 void nts_stream_sync(nts_stream *s);                 /* Cuda_Stream::CUDA_DEVICE_SYNCHRONIZE, ntsCUDAGraphOP.cu:69-75 */
 void *nts_stream_handle(nts_stream *s);              /* raw hipStream_t, for torch.cuda.ExternalStream interop */
+void nts_stream_wait_stream(nts_stream *waiter, nts_stream *waitee);
+    /* stream-ordered cross-stream dependency (hipEventRecord on waitee +
+     * hipStreamWaitEvent on waiter): how the comm stream and compute
+     * stream of the overlapped ring synchronize (the reference's
+     * PROC_OVERLAP, graph.hpp:3490-3535, is the default here) */
 
 /* Kernel-time accounting with HIP events on the launching stream (feeds
  * bench.py's roofline.achieved; replaces the reference's wall-clock

@@ -1208,6 +1208,17 @@ void nts_stream_destroy(nts_stream *s) {
 void nts_stream_sync(nts_stream *s) { NTS_CHECK(hipStreamSynchronize(s->stream)); }
 void *nts_stream_handle(nts_stream *s) { return (void *)s->stream; }
 
+void nts_stream_wait_stream(nts_stream *waiter, nts_stream *waitee) {
+  /* hipEventDestroy is stream-safe after the wait is enqueued (the wait
+   * captured the record at call time); creation cost is ~µs, called P-1
+   * times per layer */
+  hipEvent_t ev;
+  NTS_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  NTS_CHECK(hipEventRecord(ev, waitee->stream));
+  NTS_CHECK(hipStreamWaitEvent(waiter->stream, ev, 0));
+  NTS_CHECK(hipEventDestroy(ev));
+}
+
 void nts_stream_timing(nts_stream *s, int enable) { s->timing = enable != 0; }
 
 void nts_stream_timing_reset(nts_stream *s) {
