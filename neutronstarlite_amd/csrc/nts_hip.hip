@@ -578,6 +578,44 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
      * [fanout, CAP], collect survivors to LDS, then select the fanout
      * smallest among them. */
     const uint32_t cap = CAP < 4 * fanout ? CAP : 4 * fanout;
+    if (deg <= CAP && !FORCE_FALLBACK) {
+      /* every candidate fits the collection buffer: skip the threshold
+       * search entirely (its ~5-36 deg-scans cost more than selecting the
+       * fanout smallest from deg candidates directly) */
+      if (lane == 0) s_n[wib] = 0;
+      __builtin_amdgcn_wave_barrier();
+      for (uint32_t j = lane; j < deg; j += 64) {
+        const uint32_t p = atomicAdd(&s_n[wib], 1u);
+        s_key[wib][p] = k_hash_u32(seed, 0, e0 + j);
+        s_slot[wib][p] = j;
+      }
+      __builtin_amdgcn_wave_barrier();
+      const uint32_t m = s_n[wib];
+      for (uint32_t pick = 0; pick < k; ++pick) {
+        unsigned long long best = ~0ULL;
+        uint32_t bp = 0;
+        for (uint32_t p = lane; p < m; p += 64) {
+          const unsigned long long cand =
+              ((unsigned long long)s_key[wib][p] << 32) | s_slot[wib][p];
+          if (cand < best) { best = cand; bp = p; }
+        }
+#pragma unroll
+        for (int w = 32; w >= 1; w >>= 1) {
+          const unsigned long long ob = __shfl_xor(best, w, 64);
+          const uint32_t op = __shfl_xor(bp, w, 64);
+          if (ob < best) { best = ob; bp = op; }
+        }
+        if (lane == 0) {
+          out_src[(uint64_t)i * fanout + pick] =
+              row_indices[e0 + s_slot[wib][bp]];
+          s_key[wib][bp] = 0xFFFFFFFFu;  /* removed: cand becomes ~0ULL */
+          s_slot[wib][bp] = 0xFFFFFFFFu;
+        }
+        __builtin_amdgcn_wave_barrier();
+      }
+      if (lane == 0) out_cnt[i] = k;
+      continue;
+    }
     unsigned long long lo = 0, hi = 0x100000000ULL;
     unsigned long long T =
         (unsigned long long)(2.0 * fanout / deg * 4294967296.0) + 1;
@@ -656,7 +694,10 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
       if (lane == 0) {
         out_src[(uint64_t)i * fanout + pick] =
             row_indices[e0 + s_slot[wib][bp]];
-        s_key[wib][bp] = 0xFFFFFFFFu;  /* remove from candidates */
+        s_key[wib][bp] = 0xFFFFFFFFu;  /* removed: cand becomes ~0ULL so a
+                                          genuine max-key candidate can
+                                          never be shadowed */
+        s_slot[wib][bp] = 0xFFFFFFFFu;
       }
       __builtin_amdgcn_wave_barrier();
     }
