@@ -58,6 +58,9 @@ def _load():
     for fn in (lib.nts_ref_src_scatter_fwd, lib.nts_ref_src_scatter_bwd,
                lib.nts_ref_dst_aggregate_fwd, lib.nts_ref_dst_aggregate_bwd):
         fn.argtypes = [u32, u64, i64, u32p, u32p, f32p, f32p]
+    for fn in (lib.nts_ref_minibatch_forward, lib.nts_ref_minibatch_backward):
+        fn.argtypes = [u32, u32, u32, i64, u32p, u32p, u32p, u32p, u32p,
+                       u32p, f32p, f32p]
     lib.nts_ref_ok.restype = ctypes.c_int
     assert lib.nts_ref_ok() == 1
     _lib = lib
@@ -136,6 +139,31 @@ def dst_aggregate_fwd(v, e, f, col_off, rows, msg):
     _load().nts_ref_dst_aggregate_fwd(v, e, f, _u32(col_off), _u32(rows),
                                       _f32(msg), _f32(y))
     return y
+
+
+def minibatch_forward(v, f, col_off, r_i_local, dst_ids, src_ids, outdeg,
+                      indeg, x):
+    """MiniBatchFuseOp::forward (ntsMiniBatchGraphOp.hpp:71-101): sampled
+    CSC aggregation with norm-degree weights over GLOBAL degrees."""
+    n_dst, n_src = len(dst_ids), len(src_ids)
+    y = np.zeros((n_dst, f), dtype=np.float32)
+    _load().nts_ref_minibatch_forward(v, n_dst, n_src, f, _u32(col_off),
+                                      _u32(r_i_local), _u32(dst_ids),
+                                      _u32(src_ids), _u32(outdeg),
+                                      _u32(indeg), _f32(x), _f32(y))
+    return y
+
+
+def minibatch_backward(v, f, col_off, r_i_local, dst_ids, src_ids, outdeg,
+                       indeg, gy):
+    """MiniBatchFuseOp::backward (ntsMiniBatchGraphOp.hpp:102-129)."""
+    n_dst, n_src = len(dst_ids), len(src_ids)
+    gx = np.zeros((n_src, f), dtype=np.float32)
+    _load().nts_ref_minibatch_backward(v, n_dst, n_src, f, _u32(col_off),
+                                       _u32(r_i_local), _u32(dst_ids),
+                                       _u32(src_ids), _u32(outdeg),
+                                       _u32(indeg), _f32(gy), _f32(gx))
+    return gx
 
 
 def dst_aggregate_bwd(v, e, f, col_off, rows, y_grad):

@@ -28,9 +28,17 @@
 #include <vector>
 
 #include "core/ntsCPUFusedGraphOp.hpp"   /* reference source, via -I */
+/* ntsMiniBatchGraphOp.hpp includes "ntsSampler.hpp", which resolves to the
+ * includer's directory (the REAL core/ntsSampler.hpp -> FullyRepGraph/MPI
+ * machinery).  Neutralize its guard and supply the 1-rank stub surface
+ * instead; the op's loop bodies still compile from the reference tree. */
+#define NTSSAMPLER_HPP
+#include "ntsSampler.hpp"                /* stub/ntsSampler.hpp via -Istub */
+#include "core/ntsMiniBatchGraphOp.hpp"  /* reference source, via -I */
 #include "core/ntsSingleCPUGraphOp.hpp"  /* reference source, via -I */
 
 using nts::op::ForwardCPUfuseOp;
+using nts::op::MiniBatchFuseOp;
 using nts::op::SingleCPUDstAggregateOp;
 using nts::op::SingleCPUSrcScatterOp;
 
@@ -177,6 +185,73 @@ EXPORT void nts_ref_dst_aggregate_bwd(uint32_t v, uint64_t e, int64_t f,
   NtsVar gv(v, f, const_cast<float *>(y_grad));
   NtsVar mv = op.backward(gv);
   memcpy(msg_grad, mv.data(), sizeof(float) * e * f);
+}
+
+/* ---- MiniBatchFuseOp (core/ntsMiniBatchGraphOp.hpp:61-131) ----
+ * Sampled-subgraph aggregation: local CSC over n_dst sampled destinations,
+ * r_i entries are LOCAL source slots, dst_ids/src_ids map locals to global
+ * vertex ids (for the norm-degree weights). */
+static void minibatch_world(Graph<Empty> &g, StubGnnCtx &ctx, StubNts &nts,
+                            const uint32_t *outdeg, const uint32_t *indeg,
+                            uint32_t v) {
+  g.vertices = v;
+  g.out_degree_for_backward = const_cast<uint32_t *>(outdeg);
+  g.in_degree_for_backward = const_cast<uint32_t *>(indeg);
+  ctx.p_v_s = 0;
+  g.gnnctx = &ctx;
+  g.Nts = &nts;
+}
+
+EXPORT void nts_ref_minibatch_forward(uint32_t v, uint32_t n_dst,
+                                      uint32_t n_src, int64_t f,
+                                      const uint32_t *col_off,
+                                      const uint32_t *r_i_local,
+                                      const uint32_t *dst_ids,
+                                      const uint32_t *src_ids,
+                                      const uint32_t *outdeg,
+                                      const uint32_t *indeg, const float *x,
+                                      float *y) {
+  Graph<Empty> g;
+  StubGnnCtx ctx;
+  StubNts nts;
+  minibatch_world(g, ctx, nts, outdeg, indeg, v);
+  sampCSC sg;
+  sg.v_dst.assign(dst_ids, dst_ids + n_dst);
+  sg.v_src.assign(src_ids, src_ids + n_src);
+  sg.c_o.assign(col_off, col_off + n_dst + 1);
+  sg.r_indices.assign(r_i_local, r_i_local + col_off[n_dst]);
+  SampledSubgraph ssg;
+  ssg.sampled_sgs.push_back(&sg);
+  MiniBatchFuseOp op(&ssg, &g, /*layer=*/0);
+  NtsVar xv(n_src, f, const_cast<float *>(x));
+  NtsVar yv = op.forward(xv);
+  memcpy(y, yv.data(), sizeof(float) * n_dst * f);
+}
+
+EXPORT void nts_ref_minibatch_backward(uint32_t v, uint32_t n_dst,
+                                       uint32_t n_src, int64_t f,
+                                       const uint32_t *col_off,
+                                       const uint32_t *r_i_local,
+                                       const uint32_t *dst_ids,
+                                       const uint32_t *src_ids,
+                                       const uint32_t *outdeg,
+                                       const uint32_t *indeg, const float *gy,
+                                       float *gx) {
+  Graph<Empty> g;
+  StubGnnCtx ctx;
+  StubNts nts;
+  minibatch_world(g, ctx, nts, outdeg, indeg, v);
+  sampCSC sg;
+  sg.v_dst.assign(dst_ids, dst_ids + n_dst);
+  sg.v_src.assign(src_ids, src_ids + n_src);
+  sg.c_o.assign(col_off, col_off + n_dst + 1);
+  sg.r_indices.assign(r_i_local, r_i_local + col_off[n_dst]);
+  SampledSubgraph ssg;
+  ssg.sampled_sgs.push_back(&sg);
+  MiniBatchFuseOp op(&ssg, &g, 0);
+  NtsVar gv(n_dst, f, const_cast<float *>(gy));
+  NtsVar gxv = op.backward(gv);
+  memcpy(gx, gxv.data(), sizeof(float) * n_src * f);
 }
 
 EXPORT int nts_ref_ok(void) { return 1; }

@@ -83,6 +83,7 @@ struct NtsVar {
   NtsVar mm(const NtsVar &) const { abort(); }
   NtsVar operator*(const NtsVar &) const { abort(); }
   NtsVar operator-(const NtsVar &) const { abort(); }
+  NtsVar &operator[](long) { abort(); } /* get_label's per-row copy */
 };
 
 struct VertexSubset {
@@ -143,6 +144,8 @@ struct StubNts {
   NtsVar NewLeafTensor(std::vector<long> shape, torch::DeviceType) {
     return NtsVar(shape[0], shape[1]);
   }
+  NtsVar NewLeafKLongTensor(std::vector<long>) { abort(); } /* get_label
+      only; never executed through this stub */
   ValueType *getWritableBuffer(NtsVar &v, torch::DeviceType) {
     return v.data();
   }
