@@ -107,3 +107,35 @@ def test_work_items_cover_and_flag(split):
     # flags: set iff vertex has >1 item
     n_items = np.bincount(vids, minlength=v)
     assert np.all((n_items[vids] > 1) == (flags == 1))
+
+
+def test_degenerate_sizes():
+    """Edge cases the reference never guards: single-vertex graphs, an
+    all-self-loop graph, and a vertex range with zero edges must flow
+    through degrees/weights/chunks without error."""
+    # single vertex, single self loop
+    edges = np.array([[0, 0]], dtype=np.uint32)
+    outd, ind = G.degrees(edges, 1)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    ch = G.build_chunks(edges, w, np.array([0, 1], dtype=np.uint32), 0)[0]
+    assert ch.edge_size == 1 and ch.column_offset[-1] == 1
+    # all-self-loop graph
+    v = 17
+    edges = np.stack([np.arange(v, dtype=np.uint32)] * 2, axis=1)
+    outd, ind = G.degrees(edges, v)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    assert np.allclose(w, 1.0)
+    ch = G.build_chunks(edges, w, np.array([0, v], dtype=np.uint32), 0)[0]
+    import oracle
+    x = np.random.default_rng(0).normal(size=(v, 3)).astype(np.float32)
+    y = oracle.csc_forward(ch.column_offset, ch.row_indices,
+                           ch.edge_weight_forward, x, 0, v, 3)
+    assert np.allclose(y, x, rtol=1e-6)
+    # zero-edge partition range: chunks exist with empty CSC
+    edges = np.array([[0, 1]], dtype=np.uint32)
+    outd, ind = G.degrees(edges, 8)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    offs = np.array([0, 4, 8], dtype=np.uint32)
+    chunks = G.build_chunks(edges, w, offs, 1)  # rank 1 owns dst 4..8: none
+    assert all(ch.edge_size == 0 for ch in chunks)
+    assert all(ch.column_offset[-1] == 0 for ch in chunks)
