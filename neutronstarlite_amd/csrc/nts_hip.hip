@@ -556,7 +556,6 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
   constexpr uint32_t CAP = 1024;
   __shared__ uint32_t s_key[4][CAP];
   __shared__ uint32_t s_slot[4][CAP];
-  __shared__ uint32_t s_n[4];
   const uint32_t wib = threadIdx.x >> 6;
   const uint32_t lane = threadIdx.x & 63;
   const uint32_t wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
@@ -580,17 +579,16 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
     const uint32_t cap = CAP < 4 * fanout ? CAP : 4 * fanout;
     if (deg <= CAP && !FORCE_FALLBACK) {
       /* every candidate fits the collection buffer: skip the threshold
-       * search entirely (its ~5-36 deg-scans cost more than selecting the
-       * fanout smallest from deg candidates directly) */
-      if (lane == 0) s_n[wib] = 0;
-      __builtin_amdgcn_wave_barrier();
+       * search entirely and store DIRECTLY at the slot index — no LDS
+       * counter (a single atomicAdd counter serialized all 64 lanes and
+       * was the kernel's hot spot: 65 % of the sampled-step GPU time,
+       * profiles/round2) */
       for (uint32_t j = lane; j < deg; j += 64) {
-        const uint32_t p = atomicAdd(&s_n[wib], 1u);
-        s_key[wib][p] = k_hash_u32(seed, 0, e0 + j);
-        s_slot[wib][p] = j;
+        s_key[wib][j] = k_hash_u32(seed, 0, e0 + j);
+        s_slot[wib][j] = j;
       }
       __builtin_amdgcn_wave_barrier();
-      const uint32_t m = s_n[wib];
+      const uint32_t m = deg;
       for (uint32_t pick = 0; pick < k; ++pick) {
         unsigned long long best = ~0ULL;
         uint32_t bp = 0;
@@ -663,18 +661,28 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
         else { hi64 = T64; T64 = lo64 + (T64 - lo64) / 2; }
       }
     }
-    if (lane == 0) s_n[wib] = 0;
-    __builtin_amdgcn_wave_barrier();
-    for (uint32_t j = lane; j < deg; j += 64) {
-      const unsigned long long cand =
-          ((unsigned long long)k_hash_u32(seed, 0, e0 + j) << 32) | j;
-      if (cand < T64) {
-        const uint32_t p = atomicAdd(&s_n[wib], 1u);
-        if (p < CAP) { s_key[wib][p] = (uint32_t)(cand >> 32); s_slot[wib][p] = j; }
+    /* ballot-compacted collection: survivors land slot-ordered with no
+     * LDS atomics (the old per-lane atomicAdd counter serialized the
+     * wave; placement is now also deterministic by construction) */
+    uint32_t base = 0;
+    for (uint32_t j0 = 0; j0 < deg; j0 += 64) {
+      const uint32_t j = j0 + lane;
+      bool pred = false;
+      uint32_t key = 0;
+      if (j < deg) {
+        key = k_hash_u32(seed, 0, e0 + j);
+        pred = (((unsigned long long)key << 32) | j) < T64;
       }
+      const unsigned long long bmask = __ballot(pred);
+      if (pred) {
+        const uint32_t p =
+            base + (uint32_t)__popcll(bmask & ((1ULL << lane) - 1));
+        if (p < CAP) { s_key[wib][p] = key; s_slot[wib][p] = j; }
+      }
+      base += (uint32_t)__popcll(bmask);
     }
     __builtin_amdgcn_wave_barrier();
-    const uint32_t m = s_n[wib] < CAP ? s_n[wib] : CAP;
+    const uint32_t m = base < CAP ? base : CAP;
     /* selection: repeatedly pick the minimum (key, slot) — the slot in the
      * low bits makes tie-breaks deterministic in the edge slot */
     for (uint32_t pick = 0; pick < k; ++pick) {
