@@ -577,6 +577,41 @@ __global__ void k_sample_reservoir(const uint32_t *__restrict__ column_offset,
      * [fanout, CAP], collect survivors to LDS, then select the fanout
      * smallest among them. */
     const uint32_t cap = CAP < 4 * fanout ? CAP : 4 * fanout;
+    if (deg > CAP && fanout <= 64 && !FORCE_FALLBACK) {
+      /* HUB destinations: any scan-based scheme is O(deg) on ONE wave —
+       * the layer-2 frontier is degree-biased and its biggest hub
+       * (in-degree ~814k at Reddit scale) serialized the whole launch
+       * (~3 ms measured, profiles/round2).  Same contract ("min(deg,
+       * fanout) uniformly chosen slots", ntsSampler.hpp:113-166 — whose
+       * own reservoir uses rand_r()%i), delivered in O(fanout):
+       * rejection sampling without replacement, slot_p = hash(seed, d,
+       * pick | attempt<<10) % deg; duplicates against earlier picks
+       * re-roll.  Deterministic in (seed, dst, pick): no scan, no
+       * atomic order, bias <= deg/2^32 (way under the reference's
+       * rand()%i bias). */
+      const bool active = lane < fanout;
+      uint32_t attempt = 0;
+      uint32_t slot = active
+          ? k_hash_u32(seed, d, lane | (attempt << 10)) % deg : 0xFFFFFFFFu;
+      for (int round = 0; round < 100; ++round) {
+        s_slot[wib][lane] = slot;
+        __builtin_amdgcn_wave_barrier();
+        bool dup = false;
+        if (active) {
+          for (uint32_t q = 0; q < lane; ++q)
+            if (s_slot[wib][q] == slot) { dup = true; break; }
+        }
+        __builtin_amdgcn_wave_barrier();
+        if (__ballot(dup) == 0) break;
+        if (dup) {
+          ++attempt;
+          slot = k_hash_u32(seed, d, lane | (attempt << 10)) % deg;
+        }
+      }
+      if (active) out_src[(uint64_t)i * fanout + lane] = row_indices[e0 + slot];
+      if (lane == 0) out_cnt[i] = fanout;
+      continue;
+    }
     if (deg <= CAP && !FORCE_FALLBACK) {
       /* every candidate fits the collection buffer: skip the threshold
        * search entirely and store DIRECTLY at the slot index — no LDS
