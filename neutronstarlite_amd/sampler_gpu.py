@@ -61,15 +61,29 @@ def sample_layer_gpu(stream: shim.Stream, d_column_offset: torch.Tensor,
                             n, fanout, seed, out_src.data_ptr(),
                             out_cnt.data_ptr())
     cnt = out_cnt.to(torch.int64)
-    valid = (torch.arange(fanout, device=dev)[None, :] < cnt[:, None])
-    src_g = out_src.view(n, fanout)[valid].to(torch.int64) & 0xFFFFFFFF
-    # row-major mask selection keeps edges grouped by destination => the
-    # selected order IS the local CSC order
-    d_local = (torch.arange(n, device=dev)[:, None]
-               .expand(n, fanout)[valid])
     col_off = torch.zeros(n + 1, dtype=torch.int64, device=dev)
     col_off[1:] = torch.cumsum(cnt, 0)
-    src_unique, ril = torch.unique(src_g, return_inverse=True)
+    # compaction with exactly TWO host syncs (E and n_src): boolean-mask
+    # selection and torch.unique each forced an extra device->host sync per
+    # layer, which dominated the sampled step once the kernels got fast
+    e_sz = int(col_off[-1].item())                       # sync 1
+    d_local = torch.repeat_interleave(
+        torch.arange(n, device=dev), cnt, output_size=e_sz)
+    j_within = torch.arange(e_sz, device=dev) - col_off[d_local]
+    # dst-major order == the local CSC order
+    src_g = (out_src.view(-1).to(torch.int64)[d_local * fanout + j_within]
+             & 0xFFFFFFFF)
+    # sorted-unique + inverse via sort/flags (no torch.unique sync)
+    sorted_src, sort_idx = torch.sort(src_g)
+    new_flag = torch.ones(e_sz, dtype=torch.bool, device=dev)
+    if e_sz > 1:
+        new_flag[1:] = sorted_src[1:] != sorted_src[:-1]
+    ril_sorted = torch.cumsum(new_flag, 0) - 1
+    ril = torch.empty_like(ril_sorted)
+    ril[sort_idx] = ril_sorted
+    n_uniq = int(ril_sorted[-1].item()) + 1 if e_sz else 0   # sync 2
+    src_unique = torch.zeros(n_uniq, dtype=torch.int64, device=dev)
+    src_unique[ril_sorted] = sorted_src   # duplicate writes carry equal values
     dst_g = (dst_list.to(torch.int64) & 0xFFFFFFFF)[d_local]
     w = (1.0 / (torch.sqrt(d_outd[src_g].float()) *
                 torch.sqrt(d_ind[dst_g].float())))
