@@ -166,7 +166,7 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                          "workload instead of --graph")
     ap.add_argument("--model", default="gcn",
                     choices=["gcn", "gat", "gcn-layer", "gcn-sample",
-                             "gcn-train"],
+                             "gcn-sample-train", "gcn-train"],
                     help="gcn = fused norm-degree aggregation (configs #2-4);"
                          " gat = attention-weighted layer with edge softmax "
                          "(config #5, single GPU, --feat 128); gcn-layer = "
@@ -299,7 +299,7 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         dchunks = chunks
         engine = _test_engine_factory()
     engine.stream.timing(True)
-    if _test_engine_factory is None and args.model != "gcn-sample":
+    if _test_engine_factory is None and not args.model.startswith("gcn-sample"):
         # chunks are static for the full-batch modes -> work-item reuse is
         # safe (sampled subgraphs change every step, so gcn-sample keeps
         # the rebuild-per-call default)
@@ -363,11 +363,16 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
             loss.backward()
             opt.step()
             return loss
-    elif args.model == "gcn-sample":
+    elif args.model in ("gcn-sample", "gcn-sample-train"):
         # mini-batch step: sample a 2-layer subgraph on device
         # (reservoir kernel + torch compaction), then aggregate innermost ->
         # outermost with the same gather kernels.  Value still counts
         # aggregated edges (the sampled edges of both layers, fwd+bwd).
+        # gcn-sample-train is the END-TO-END training step:
+        # sampled feature gather from the resident feature matrix + 2-layer
+        # GCN (agg -> W0 -> relu -> agg -> W1 -> nll) through the autograd
+        # bridge + Adam — the reference's GCN_CPU_SAMPLE per-epoch loop
+        # (toolkits/GCN_CPU_SAMPLE.hpp:195-260) on GPU-resident sampling.
         assert not distributed, "sampled bench is single-GPU here"
         from neutronstarlite_amd.ops import MiniBatchFuseOp, _u32_cuda
         from neutronstarlite_amd.sampler_gpu import sample_subgraph_gpu
@@ -388,27 +393,64 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
         gen = np.random.default_rng(42)
         target_pool = gen.permutation(v).astype(np.int32)
 
-        def step():
-            i = step_idx[0]
-            step_idx[0] += 1
-            lo_t = (i * args.batch_size) % max(1, v - args.batch_size)
-            targets = torch.from_numpy(
-                target_pool[lo_t:lo_t + args.batch_size]).to(dev)
-            layers = sample_subgraph_gpu(engine.stream, d_coff, d_rows,
-                                         targets, fanouts, d_outd, d_ind,
-                                         seed=1000 + i)
-            h = torch.randn(layers[-1].n_src, f, device=dev)
-            for ly in reversed(layers):
-                op = MiniBatchFuseOp(ly, dev, engine)
-                h = op.forward(h.contiguous())
-                sampled_edges[0] += ly.e_size
-            gy = torch.randn(layers[0].n_dst, f, device=dev)
-            g = gy
-            for ly in layers:
-                op = MiniBatchFuseOp(ly, dev, engine)
-                g = op.backward(g.contiguous())
-                sampled_edges[0] += ly.e_size
-            return h, g
+        if args.model == "gcn-sample-train":
+            from neutronstarlite_amd.ops import minibatch_aggregate
+            f1, ncls = args.feat_out, 41  # LAYERS 602-128-41 convention
+            gw = torch.Generator().manual_seed(7)
+            W0 = ((torch.rand(f, f1, generator=gw) * 0.2 - 0.1)
+                  .to(dev).requires_grad_(True))
+            W1 = ((torch.rand(f1, ncls, generator=gw) * 0.2 - 0.1)
+                  .to(dev).requires_grad_(True))
+            opt = torch.optim.Adam([W0, W1], lr=1e-2)
+            label_pool = torch.from_numpy(
+                gen.integers(0, ncls, v).astype(np.int64)).to(dev)
+
+            def step():
+                i = step_idx[0]
+                step_idx[0] += 1
+                lo_t = (i * args.batch_size) % max(1, v - args.batch_size)
+                targets = torch.from_numpy(
+                    target_pool[lo_t:lo_t + args.batch_size]).to(dev)
+                layers = sample_subgraph_gpu(engine.stream, d_coff, d_rows,
+                                             targets, fanouts, d_outd, d_ind,
+                                             seed=1000 + i)
+                ops_ = [MiniBatchFuseOp(ly, dev, engine) for ly in layers]
+                # sampled feature gather from the RESIDENT matrix
+                x_s = x.index_select(0, layers[-1].src)
+                opt.zero_grad(set_to_none=True)
+                h = minibatch_aggregate(x_s, ops_[-1])
+                h = torch.relu(h @ W0)
+                h = minibatch_aggregate(h, ops_[0])
+                out = torch.log_softmax(h @ W1, 1)
+                loss = torch.nn.functional.nll_loss(
+                    out, label_pool[targets.to(torch.int64)])
+                loss.backward()
+                opt.step()
+                for ly in layers:
+                    sampled_edges[0] += 2 * ly.e_size  # fwd + bwd
+                return loss
+        else:
+            def step():
+                i = step_idx[0]
+                step_idx[0] += 1
+                lo_t = (i * args.batch_size) % max(1, v - args.batch_size)
+                targets = torch.from_numpy(
+                    target_pool[lo_t:lo_t + args.batch_size]).to(dev)
+                layers = sample_subgraph_gpu(engine.stream, d_coff, d_rows,
+                                             targets, fanouts, d_outd, d_ind,
+                                             seed=1000 + i)
+                h = torch.randn(layers[-1].n_src, f, device=dev)
+                for ly in reversed(layers):
+                    op = MiniBatchFuseOp(ly, dev, engine)
+                    h = op.forward(h.contiguous())
+                    sampled_edges[0] += ly.e_size
+                gy = torch.randn(layers[0].n_dst, f, device=dev)
+                g = gy
+                for ly in layers:
+                    op = MiniBatchFuseOp(ly, dev, engine)
+                    g = op.backward(g.contiguous())
+                    sampled_edges[0] += ly.e_size
+                return h, g
     elif args.model == "gcn-layer":
         # the layer's dense projection (x·W, the reference's
         # P[layer]->forward at NtsScheduler.hpp:737-740) ordered BEFORE the
@@ -514,7 +556,7 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
     if args.model == "gcn-train":
         # one epoch = 2 layers x (fwd + bwd) aggregation
         value = args.steps * 4.0 * e_total / elapsed
-    elif args.model == "gcn-sample":
+    elif args.model.startswith("gcn-sample"):
         # count the edges actually sampled+aggregated during the timed steps
         # (warmup's share removed via the step counter)
         value = sampled_edges[0] * (args.steps /
@@ -552,6 +594,12 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                        f"{args.batch_size} fanout={args.fanout} + sampled "
                        "aggregation fwd+bwd (SURVEY 8f-3)"
                        if args.model == "gcn-sample" else
+                       f"mini-batch GCN TRAINING step: GPU-resident "
+                       f"sampling batch={args.batch_size} fanout="
+                       f"{args.fanout} + feature gather + 2-layer "
+                       f"({f}-{args.feat_out}-41) agg+mm fwd/bwd + Adam "
+                       "(GCN_CPU_SAMPLE loop on device)"
+                       if args.model == "gcn-sample-train" else
                        "GCN-layer aggregation fwd+bwd"
                        + (" (BASELINE config #2)" if args.graph == "reddit"
                           else ""))),
@@ -560,7 +608,8 @@ def main(argv=None, _test_engine_factory=None, _test_backend=None,
                 "parallelism": f"graph-partitioned dp{n}",
                 "edges_per_step": 2 * e_total,
             },
-            "roofline": (None if args.model == "gcn-sample" else roofline),
+            "roofline": (None if args.model.startswith("gcn-sample")
+                         else roofline),
             "cpu_baseline": cpu_baseline,
         }
         print(json.dumps(out), flush=True)

@@ -118,6 +118,27 @@ def aggregate(x: torch.Tensor, dchunk: "DeviceChunk",
     return _AggregateFn.apply(x, dchunk, engine)
 
 
+class _MiniBatchAggFn(torch.autograd.Function):
+    """torch.autograd bridge over one sampled layer's MiniBatchFuseOp, so
+    whole mini-batch training loops (the reference's GCN_CPU_SAMPLE per-layer
+    chain, toolkits/GCN_CPU_SAMPLE.hpp:214 + ntsMiniBatchGraphOp.hpp:61-131)
+    are plain torch code."""
+
+    @staticmethod
+    def forward(ctx, x, op):
+        ctx.op = op
+        return op.forward(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, grad_y):
+        return ctx.op.backward(grad_y.contiguous()), None
+
+
+def minibatch_aggregate(x: torch.Tensor, op: "MiniBatchFuseOp") -> torch.Tensor:
+    """Autograd-aware sampled-subgraph aggregation."""
+    return _MiniBatchAggFn.apply(x, op)
+
+
 class MiniBatchFuseOp:
     """MiniBatchFuseOp equivalent (core/ntsMiniBatchGraphOp.hpp:61-131):
     the same aggregation arithmetic on one sampled layer's compacted
