@@ -36,10 +36,11 @@ def torch_gat_reference(h, a_src, a_dst, dst_of_edge, src_of_edge, v, slope):
     return y
 
 
-@pytest.mark.parametrize("f", [32, 33, 128])
+@pytest.mark.parametrize("f", [32, 33, 128, 256])
 def test_gat_layer_forward_backward(f):
-    """f=32/128 run the fused CSR-gather+dot path; f=33 (odd) exercises the
-    unfused fallback (rc=0 -> separate edge-dot) through the same layer."""
+    """f=32/128 (G=32) and f=256 (G=64) run the fused CSR-gather+dot path;
+    f=33 (odd) exercises the unfused fallback (rc=0 -> separate edge-dot)
+    through the same layer."""
     from neutronstarlite_amd.gat import GATLayer
     dev = torch.device("cuda:0")
     v, e, slope = 1200, 20000, 0.2
