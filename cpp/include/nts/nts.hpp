@@ -100,7 +100,10 @@ struct PartitionedGraph {
    * fills/copies and our kernels must share one stream order. */
   PartitionedGraph() {
     stream = nts_stream_wrap(nullptr);
-    comm_stream = nts_stream_create();  /* non-blocking side stream */
+    /* the side stream needs a device; CPU-only host-logic tests
+     * (cpp/host_unit_check.cpp) construct this object without one */
+    if (nts_device_count() > 0)
+      comm_stream = nts_stream_create();  /* non-blocking side stream */
   }
   ~PartitionedGraph() {
     if (stream) nts_stream_destroy(stream);

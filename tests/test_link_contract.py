@@ -29,3 +29,18 @@ def test_flagship_loop_compiles_and_links():
     r = subprocess.run([BIN], capture_output=True, text=True, timeout=300,
                        env=_env())
     assert r.returncode == 0, f"gcn_link_check failed:\n{r.stdout}\n{r.stderr}"
+
+
+def test_host_layer_units_cpu():
+    """CPU-runnable unit checks of the C++ host layer: Parameter's
+    hand-rolled Adam + decay bookkeeping vs a double-precision
+    recomputation, and the NtsContext tape's graph-op/NN-segment
+    interleaving with analytic gradients (cpp/host_unit_check.cpp)."""
+    bin_ = os.path.join(REPO, "cpp", "build", "host_unit_check")
+    if not os.path.exists(bin_):
+        import __graft_entry__
+        __graft_entry__._build_cpp()
+    r = subprocess.run([bin_], capture_output=True, text=True, timeout=300,
+                       env=_env())
+    assert r.returncode == 0, f"host_unit_check:\n{r.stdout}\n{r.stderr}"
+    assert "host_unit_check ok" in r.stdout
