@@ -121,6 +121,44 @@ def test_edge_decomposed_ops_bit_equal():
         assert np.array_equal(mg, mg_ref), name
 
 
+def test_property_random_graphs_bit_equal():
+    """Hypothesis-style sweep: arbitrary small graphs (duplicate edges,
+    self loops, isolated vertices, hub columns, empty columns) must keep
+    the oracle bit-equal to the reference-executed ForwardCPUfuseOp on
+    both directions."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(2, 120), st.integers(0, 600),
+           st.integers(0, 2**31 - 1), st.sampled_from([1, 3, 8]))
+    def prop(v, e, seed, f):
+        rng = np.random.default_rng(seed)
+        parts = [np.stack([rng.integers(0, v, e), rng.integers(0, v, e)],
+                          axis=1).astype(np.uint32)]
+        # self loops + a hub column (every vertex -> vertex 0)
+        parts.append(np.stack([np.arange(v, dtype=np.uint32)] * 2, axis=1))
+        parts.append(np.stack([np.arange(v, dtype=np.uint32),
+                               np.zeros(v, dtype=np.uint32)], axis=1))
+        edges = np.concatenate(parts, axis=0)
+        ch, outd, ind, _ = _world(edges, v)
+        x = rng.uniform(-1, 1, size=(v, f)).astype(np.float32)
+        g = rng.uniform(-1, 1, size=(v, f)).astype(np.float32)
+        y_ref = ref.fused_forward(v, f, ch.column_offset, ch.row_indices,
+                                  ch.row_offset, ch.column_indices, outd,
+                                  ind, x)
+        y = oracle.csc_forward(ch.column_offset, ch.row_indices,
+                               ch.edge_weight_forward, x, 0, v, f)
+        assert np.array_equal(y, y_ref)
+        gx_ref = ref.fused_backward(v, f, ch.column_offset, ch.row_indices,
+                                    ch.row_offset, ch.column_indices, outd,
+                                    ind, g)
+        gx = oracle.csr_backward(ch.row_offset, ch.column_indices,
+                                 ch.edge_weight_backward, g, 0, v, f)
+        assert np.array_equal(gx, gx_ref)
+
+    prop()
+
+
 def test_minibatch_fuse_op_bit_equal():
     """Our sampled-subgraph aggregation semantics (compacted CSC + global
     norm-degree weights, sampler.py/oracle) == MiniBatchFuseOp compiled
