@@ -61,6 +61,15 @@ def _load():
     for fn in (lib.nts_ref_minibatch_forward, lib.nts_ref_minibatch_backward):
         fn.argtypes = [u32, u32, u32, i64, u32p, u32p, u32p, u32p, u32p,
                        u32p, f32p, f32p]
+    lib.nts_ref_dist_mirror_index.argtypes = [u32, u32, u32p, u32p, u32p]
+    lib.nts_ref_dist_mirror_index.restype = u32
+    for fn in (lib.nts_ref_dist_get_dep_nbr_fwd,
+               lib.nts_ref_dist_get_dep_nbr_bwd,
+               lib.nts_ref_dist_scatter_src_fwd,
+               lib.nts_ref_dist_scatter_src_bwd,
+               lib.nts_ref_dist_aggregate_dst_fwd,
+               lib.nts_ref_dist_aggregate_dst_bwd):
+        fn.argtypes = [u32, u32, i64, u32p, u32p, f32p, f32p]
     lib.nts_ref_ok.restype = ctypes.c_int
     assert lib.nts_ref_ok() == 1
     _lib = lib
@@ -131,6 +140,62 @@ def src_scatter_bwd(v, e, f, col_off, rows, msg_grad):
     _load().nts_ref_src_scatter_bwd(v, e, f, _u32(col_off), _u32(rows),
                                     _f32(msg_grad), _f32(gx))
     return gx
+
+
+def dist_mirror_index(v, col_off, rows):
+    """generateMirrorIndex (PartitionedGraph.hpp:295-305), reference
+    semantics: prefix sum; returns (MirrorIndex[v+1], owned_mirrors)."""
+    e = int(col_off[-1])
+    out = np.zeros(v + 1, dtype=np.uint32)
+    n = _load().nts_ref_dist_mirror_index(v, e, _u32(col_off), _u32(rows),
+                                          _u32(out))
+    return out, int(n)
+
+
+def _dist(name, v, f, col_off, rows, inp, out_rows):
+    e = int(col_off[-1])
+    out = np.zeros((out_rows, f), dtype=np.float32)
+    getattr(_load(), name)(v, e, f, _u32(col_off), _u32(rows), _f32(inp),
+                           _f32(out))
+    return out
+
+
+def dist_get_dep_nbr_fwd(v, f, col_off, rows, x, n_mirrors):
+    """DistGetDepNbrOp::forward at 1 rank (ntsDistCPUGraphOp.hpp:42-83)."""
+    return _dist("nts_ref_dist_get_dep_nbr_fwd", v, f, col_off, rows, x,
+                 n_mirrors)
+
+
+def dist_get_dep_nbr_bwd(v, f, col_off, rows, mirror_grad):
+    """DistGetDepNbrOp::backward (ntsDistCPUGraphOp.hpp:85-124)."""
+    return _dist("nts_ref_dist_get_dep_nbr_bwd", v, f, col_off, rows,
+                 mirror_grad, v)
+
+
+def dist_scatter_src_fwd(v, f, col_off, rows, mirror):
+    """DistScatterSrc::forward (ntsDistCPUGraphOp.hpp:135-159)."""
+    e = int(col_off[-1])
+    return _dist("nts_ref_dist_scatter_src_fwd", v, f, col_off, rows,
+                 mirror, e)
+
+
+def dist_scatter_src_bwd(v, f, col_off, rows, msg_grad, n_mirrors):
+    """DistScatterSrc::backward (ntsDistCPUGraphOp.hpp:161-184)."""
+    return _dist("nts_ref_dist_scatter_src_bwd", v, f, col_off, rows,
+                 msg_grad, n_mirrors)
+
+
+def dist_aggregate_dst_fwd(v, f, col_off, rows, msg):
+    """DistAggregateDst::forward (ntsDistCPUGraphOp.hpp:251-277)."""
+    return _dist("nts_ref_dist_aggregate_dst_fwd", v, f, col_off, rows,
+                 msg, v)
+
+
+def dist_aggregate_dst_bwd(v, f, col_off, rows, y_grad):
+    """DistAggregateDst::backward (ntsDistCPUGraphOp.hpp:279-305)."""
+    e = int(col_off[-1])
+    return _dist("nts_ref_dist_aggregate_dst_bwd", v, f, col_off, rows,
+                 y_grad, e)
 
 
 def dst_aggregate_fwd(v, e, f, col_off, rows, msg):

@@ -35,6 +35,7 @@
 #define NTSSAMPLER_HPP
 #include "ntsSampler.hpp"                /* stub/ntsSampler.hpp via -Istub */
 #include "core/ntsMiniBatchGraphOp.hpp"  /* reference source, via -I */
+#include "core/ntsDistCPUGraphOp.hpp"    /* reference source, via -I */
 #include "core/ntsSingleCPUGraphOp.hpp"  /* reference source, via -I */
 
 using nts::op::ForwardCPUfuseOp;
@@ -252,6 +253,151 @@ EXPORT void nts_ref_minibatch_backward(uint32_t v, uint32_t n_dst,
   NtsVar gv(n_dst, f, const_cast<float *>(gy));
   NtsVar gxv = op.backward(gv);
   memcpy(gx, gxv.data(), sizeof(float) * n_src * f);
+}
+
+/* ---- dist-GAT mirror ops (core/ntsDistCPUGraphOp.hpp) ----
+ * The whole-dst-range CSC + compressed MirrorIndex machinery that
+ * neutronstarlite_amd/dist_gat.py restates: DistGetDepNbrOp's
+ * master->mirror gather / mirror->master grad return (:34-126) and the
+ * MirrorIndex-indirected per-edge scatter/aggregate (:127-310), run at
+ * 1 rank.  MirrorIndex is built HERE exactly as generateMirrorIndex does
+ * (PartitionedGraph.hpp:295-305) from the chunk's row_indices. */
+namespace {
+struct DistWorld {
+  Graph<Empty> g;
+  PartitionedGraph pg;
+  StubGnnCtx ctx;
+  StubRtmInfo rtm;
+  StubNts nts;
+  StubComm comm;
+  VertexSubset active;
+  VertexId offs[2];
+  std::vector<VertexId> mirror_index;
+  CSC_segment_pinned chunk; /* carries the src-active filter the real
+                               dep-neighbor driver applies */
+
+  DistWorld(uint32_t v, uint32_t e, const uint32_t *col_off,
+            const uint32_t *rows) {
+    g.vertices = v;
+    g.edges = e;
+    offs[0] = 0;
+    offs[1] = v;
+    g.partition_offset = offs;
+    g.local_partition_offset = offs;
+    ctx.p_v_s = 0;
+    ctx.p_v_e = v;
+    ctx.l_v_num = v;
+    ctx.l_e_num = e;
+    g.gnnctx = &ctx;
+    rtm.lock_free = false;
+    g.rtminfo = &rtm;
+    g.Nts = &nts;
+    comm.g = &g;
+    g.NtsComm = &comm;
+    pg.graph_ = &g;
+    pg.owned_vertices = v;
+    pg.owned_edges = e;
+    pg.column_offset = const_cast<uint32_t *>(col_off);
+    pg.row_indices = const_cast<uint32_t *>(rows);
+    /* generateMirrorIndex (PartitionedGraph.hpp:295-305), verbatim
+     * semantics: prefix sum over "appears as src of an owned edge" */
+    mirror_index.assign(v + 1, 0);
+    for (VertexId i = 0; i < e; i++) mirror_index[rows[i] + 1] = 1;
+    for (VertexId i = 0; i < v; i++) mirror_index[i + 1] += mirror_index[i];
+    pg.owned_mirrors = mirror_index[v];
+    pg.MirrorIndex = mirror_index.data();
+    chunk.mirror_prefix = mirror_index.data();
+    pg.graph_chunks.push_back(&chunk);
+    active.start = 0;
+    active.end = v;
+  }
+};
+}  // namespace
+
+EXPORT uint32_t nts_ref_dist_mirror_index(uint32_t v, uint32_t e,
+                                          const uint32_t *col_off,
+                                          const uint32_t *rows,
+                                          uint32_t *out_index) {
+  DistWorld w(v, e, col_off, rows);
+  memcpy(out_index, w.pg.MirrorIndex, sizeof(uint32_t) * (v + 1));
+  return w.pg.owned_mirrors;
+}
+
+EXPORT void nts_ref_dist_get_dep_nbr_fwd(uint32_t v, uint32_t e, int64_t f,
+                                         const uint32_t *col_off,
+                                         const uint32_t *rows, const float *x,
+                                         float *mirror_out) {
+  DistWorld w(v, e, col_off, rows);
+  nts::op::DistGetDepNbrOp op(&w.pg, &w.active);
+  NtsVar xv(v, f, const_cast<float *>(x));
+  NtsVar mv = op.forward(xv);
+  memcpy(mirror_out, mv.data(), sizeof(float) * w.pg.owned_mirrors * f);
+}
+
+EXPORT void nts_ref_dist_get_dep_nbr_bwd(uint32_t v, uint32_t e, int64_t f,
+                                         const uint32_t *col_off,
+                                         const uint32_t *rows,
+                                         const float *mirror_grad,
+                                         float *x_grad) {
+  DistWorld w(v, e, col_off, rows);
+  nts::op::DistGetDepNbrOp op(&w.pg, &w.active);
+  /* harness guard: the op's backward reads row MirrorIndex[src] for EVERY
+   * src, and vertices past the last mirror index one row past
+   * owned_mirrors (a reference quirk — those vertices' grads come from an
+   * out-of-bounds read there).  One zero guard row keeps the harness
+   * well-defined: mirror-less vertices receive exactly 0, which is also
+   * our implementation's semantics. */
+  NtsVar gv(w.pg.owned_mirrors + 1, f);
+  memcpy(gv.data(), mirror_grad,
+         sizeof(float) * w.pg.owned_mirrors * f);
+  NtsVar xv = op.backward(gv);
+  memcpy(x_grad, xv.data(), sizeof(float) * v * f);
+}
+
+EXPORT void nts_ref_dist_scatter_src_fwd(uint32_t v, uint32_t e, int64_t f,
+                                         const uint32_t *col_off,
+                                         const uint32_t *rows,
+                                         const float *mirror, float *msg) {
+  DistWorld w(v, e, col_off, rows);
+  nts::op::DistScatterSrc op(&w.pg, &w.active);
+  NtsVar mv(w.pg.owned_mirrors, f, const_cast<float *>(mirror));
+  NtsVar out = op.forward(mv);
+  memcpy(msg, out.data(), sizeof(float) * e * f);
+}
+
+EXPORT void nts_ref_dist_scatter_src_bwd(uint32_t v, uint32_t e, int64_t f,
+                                         const uint32_t *col_off,
+                                         const uint32_t *rows,
+                                         const float *msg_grad,
+                                         float *mirror_grad) {
+  DistWorld w(v, e, col_off, rows);
+  nts::op::DistScatterSrc op(&w.pg, &w.active);
+  NtsVar gv(e, f, const_cast<float *>(msg_grad));
+  NtsVar out = op.backward(gv);
+  memcpy(mirror_grad, out.data(), sizeof(float) * w.pg.owned_mirrors * f);
+}
+
+EXPORT void nts_ref_dist_aggregate_dst_fwd(uint32_t v, uint32_t e, int64_t f,
+                                           const uint32_t *col_off,
+                                           const uint32_t *rows,
+                                           const float *msg, float *y) {
+  DistWorld w(v, e, col_off, rows);
+  nts::op::DistAggregateDst op(&w.pg, &w.active);
+  NtsVar mv(e, f, const_cast<float *>(msg));
+  NtsVar out = op.forward(mv);
+  memcpy(y, out.data(), sizeof(float) * v * f);
+}
+
+EXPORT void nts_ref_dist_aggregate_dst_bwd(uint32_t v, uint32_t e, int64_t f,
+                                           const uint32_t *col_off,
+                                           const uint32_t *rows,
+                                           const float *y_grad,
+                                           float *msg_grad) {
+  DistWorld w(v, e, col_off, rows);
+  nts::op::DistAggregateDst op(&w.pg, &w.active);
+  NtsVar gv(v, f, const_cast<float *>(y_grad));
+  NtsVar out = op.backward(gv);
+  memcpy(msg_grad, out.data(), sizeof(float) * e * f);
 }
 
 EXPORT int nts_ref_ok(void) { return 1; }

@@ -99,7 +99,14 @@ struct CSC_segment_pinned {
   VertexId *column_indices = nullptr; /* [E] */
   VertexId *forward_multisocket_message_index = nullptr;
   BackVertexIndex *backward_multisocket_message_index = nullptr;
-  bool src_get_active(VertexId) { return true; }
+  /* when set, src_get_active answers "is v a source of an owned edge"
+   * from the MirrorIndex prefix sum — the filter the real dep-neighbor
+   * driver applies before slotting a received record
+   * (core/graph.hpp:2907-2911) */
+  const VertexId *mirror_prefix = nullptr;
+  bool src_get_active(VertexId v) {
+    return mirror_prefix ? mirror_prefix[v + 1] > mirror_prefix[v] : true;
+  }
   bool get_forward_active(VertexId) { return true; }
 };
 
@@ -235,6 +242,33 @@ public:
       VertexId vid;
       memcpy(&vid, rec, sizeof(VertexId));
       per_msg(vid, (ValueType *)(rec + sizeof(VertexId)));
+    }
+  }
+
+  /* Dep-neighbor gather driver (get_from_dep_neighbor_mutisockets,
+   * core/graph.hpp surface) at 1 rank: every master emits its row
+   * (ascending), then the slot lambda consumes each record in arrival
+   * (= emit) order — the ascending order is load-bearing for the
+   * reference's MirrorIndex clobber-then-correct behavior on non-mirror
+   * vertices. */
+  template <typename R, typename M, typename SIG, typename SLOT>
+  void get_from_dep_neighbor_mutisockets(SIG sparse_signal, SLOT sparse_slot,
+                                         std::vector<CSC_segment_pinned *> &subgraphs,
+                                         int feature_size,
+                                         VertexSubset *active) {
+    (void)active;
+    begin_records(feature_size);
+    for (VertexId src = 0; src < vertices; src++)
+      sparse_signal(src, /*current_send_partition=*/0);
+    size_t stride = sizeofM<ValueType>(feature_size);
+    for (VertexId k = 0; k < rec_count; k++) {
+      char *rec = rec_bytes.data() + (size_t)k * stride;
+      VertexId vid;
+      memcpy(&vid, rec, sizeof(VertexId));
+      /* the real driver slots only records whose vertex is a source of an
+       * owned edge (graph.hpp:2907-2911) */
+      if (subgraphs.empty() || subgraphs[0]->src_get_active(vid))
+        sparse_slot(vid, (ValueType *)(rec + sizeof(VertexId)), /*recv_id=*/0);
     }
   }
 

@@ -192,6 +192,63 @@ def test_minibatch_fuse_op_bit_equal():
     assert np.array_equal(gx, gx_ref)
 
 
+def test_dist_mirror_machinery_bit_equal():
+    """The dist-GAT mirror machinery our dist_gat.py restates, pinned to
+    reference-executed code (ntsDistCPUGraphOp.hpp at 1 rank):
+    generateMirrorIndex's numbering, DistGetDepNbrOp's master->mirror
+    gather and mirror->master grad return, and the MirrorIndex-indirected
+    DistScatterSrc / DistAggregateDst — all bit-equal to our
+    structures/oracle ops on the same inputs."""
+    from neutronstarlite_amd.dist_gat import build_dep_graph
+
+    v, e, f = 600, 7000, 6
+    edges = G.rmat_edges(v, e, seed=17)
+    outd, ind = G.degrees(edges, v)
+    w = G.norm_weights(edges[:, 0], edges[:, 1], outd, ind)
+    dg = build_dep_graph(edges, w, np.array([0, v], dtype=np.uint32), 0, v)
+    col_off, rows = dg.column_offset, dg.row_indices
+
+    # (a) MirrorIndex numbering == our compressed mirror index
+    mi_ref, n_mirrors = ref.dist_mirror_index(v, col_off, rows)
+    assert n_mirrors == dg.n_mirrors
+    assert np.array_equal(mi_ref[dg.mirrors.astype(np.int64)],
+                          dg.mirror_index[dg.mirrors.astype(np.int64)])
+
+    rng = np.random.default_rng(23)
+    x = rng.uniform(-1, 1, size=(v, f)).astype(np.float32)
+    # (b) dep-neighbor gather: mirror matrix rows are exactly x[mirrors]
+    mirror_ref = ref.dist_get_dep_nbr_fwd(v, f, col_off, rows, x, n_mirrors)
+    assert np.array_equal(mirror_ref, x[dg.mirrors.astype(np.int64)])
+    # (c) grad return: only mirrors receive their slot's grad
+    mg = rng.uniform(-1, 1, size=(n_mirrors, f)).astype(np.float32)
+    gx_ref = ref.dist_get_dep_nbr_bwd(v, f, col_off, rows, mg)
+    expect = np.zeros((v, f), np.float32)
+    expect[dg.mirrors.astype(np.int64)] = mg
+    assert np.array_equal(gx_ref, expect)
+    # (d) MirrorIndex-indirected per-edge scatter == oracle with our index
+    msg_ref = ref.dist_scatter_src_fwd(v, f, col_off, rows, mirror_ref)
+    msg = oracle.scatter_src_to_msg(
+        np.zeros((int(col_off[-1]), f), np.float32), mirror_ref, rows,
+        col_off, dg.mirror_index, v, f)
+    assert np.array_equal(msg_ref, msg)
+    mgr_ref = ref.dist_scatter_src_bwd(v, f, col_off, rows, msg_ref,
+                                       n_mirrors)
+    mgr = oracle.gather_msg_to_src(np.zeros((n_mirrors, f), np.float32),
+                                   msg_ref, rows, col_off, dg.mirror_index,
+                                   v, f)
+    assert np.array_equal(mgr_ref, mgr)
+    # (e) edge->dst reduce + its broadcast adjoint
+    y_ref = ref.dist_aggregate_dst_fwd(v, f, col_off, rows, msg_ref)
+    y = oracle.gather_msg_to_dst(np.zeros((v, f), np.float32), msg_ref,
+                                 col_off, v, f)
+    assert np.array_equal(y_ref, y)
+    gy = rng.uniform(-1, 1, size=(v, f)).astype(np.float32)
+    mgd_ref = ref.dist_aggregate_dst_bwd(v, f, col_off, rows, gy)
+    mgd = oracle.scatter_grad_back_to_msg(
+        gy, np.zeros((int(col_off[-1]), f), np.float32), col_off, v, f)
+    assert np.array_equal(mgd_ref, mgd)
+
+
 def test_reference_src_scatter_backward_bug_documented():
     """The reference's SingleCPUSrcScatterOp::backward swaps nts_acc's
     arguments (ntsSingleCPUGraphOp.hpp:138-141): it accumulates the zeroed
