@@ -75,12 +75,14 @@ def _worker(rank, world, port, flags, q):
 
 
 @pytest.mark.timeout(420)
-@pytest.mark.parametrize("flags", [[], ["--mirror-filtered"]])
-def test_bench_orchestration_world3(flags):
-    world = 3
+@pytest.mark.parametrize("world,flags", [(3, []), (3, ["--mirror-filtered"]),
+                                         (4, []), (4, ["--mirror-filtered"])])
+def test_bench_orchestration(world, flags):
+    # world 4 runs the 8-GPU SCALE shape's ring structure (3 exchange steps
+    # per direction, pipelined backward double-buffering) under gloo
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29611 + len(flags)
+    port = 29611 + len(flags) + world * 7
     procs = [ctx.Process(target=_worker, args=(r, world, port, flags, q))
              for r in range(world)]
     for p in procs:
@@ -96,7 +98,7 @@ def test_bench_orchestration_world3(flags):
     # exactly rank 0 prints exactly one JSON line with the contract fields
     lines = [l for l in outs[0].splitlines() if l.startswith("{")]
     assert len(lines) == 1
-    for r in (1, 2):
+    for r in range(1, world):
         assert not [l for l in outs[r].splitlines() if l.startswith("{")]
     d = json.loads(lines[0])
     assert d["metric"] == "aggregated_edges_per_sec"
